@@ -1,0 +1,488 @@
+// net.cc — Net facade: device table, connection setup, request API.
+//
+// Equivalent surface to the reference's `trait Net` (src/interface.rs:34-74)
+// + BASIC backend connection logic (nthread_per_socket_backend.rs:259-522),
+// rebuilt on the nonblocking contract of ncclNet_v6 (see transport.h).
+
+#include <arpa/inet.h>
+#include <errno.h>
+#include <fcntl.h>
+#include <netinet/tcp.h>
+#include <poll.h>
+#include <string.h>
+#include <sys/socket.h>
+#include <unistd.h>
+
+#include <chrono>
+#include <random>
+
+#include "baguanet/log.h"
+#include "staging.h"
+#include "transport.h"
+
+namespace baguanet {
+
+static int set_nonblock2(int fd) {
+  int fl = fcntl(fd, F_GETFL, 0);
+  return fl < 0 ? -1 : fcntl(fd, F_SETFL, fl | O_NONBLOCK);
+}
+
+static void tune_socket2(int fd) {
+  const Config& cfg = Config::get();
+  int one = 1;
+  setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
+  if (cfg.sockbuf > 0) {
+    setsockopt(fd, SOL_SOCKET, SO_SNDBUF, &cfg.sockbuf, sizeof(cfg.sockbuf));
+    setsockopt(fd, SOL_SOCKET, SO_RCVBUF, &cfg.sockbuf, sizeof(cfg.sockbuf));
+  }
+}
+
+// Tagged request pointers: requests are 8-byte aligned, so the low bits
+// distinguish send from recv in test().
+static void* tag_send(SendRequest* r) { return (void*)((uintptr_t)r | 1); }
+static void* tag_recv(RecvRequest* r) { return (void*)((uintptr_t)r | 2); }
+
+Net& Net::get() {
+  static Net n;
+  return n;
+}
+
+Net::Net() {
+  ptr_support_ = NCCL_PTR_HOST;
+  if (staging_available()) ptr_support_ |= NCCL_PTR_CUDA;
+  auto ifs = find_interfaces();
+  devs_.reserve(ifs.size());
+  std::hash<std::string> h;
+  for (auto& nif : ifs) {
+    NetDevice d;
+    d.nif = nif;
+    d.name_str = nif.name;
+    devs_.push_back(std::move(d));
+  }
+  for (auto& d : devs_) {
+    d.props.name = const_cast<char*>(d.name_str.c_str());
+    d.props.pciPath = const_cast<char*>(d.nif.pci_path.c_str());
+    d.props.guid = (uint64_t)h(d.name_str);
+    d.props.ptrSupport = ptr_support_;
+    d.props.speed = d.nif.speed_mbps;
+    d.props.port = 0;
+    d.props.latency = 0.0f;
+    d.props.maxComms = 65536;  // reference advertised 65536 (nthread:100)
+    d.props.maxRecvs = 1;
+    BNET_INFO("baguanet device %s speed %d Mbps pci %s ptrSupport 0x%x",
+              d.name_str.c_str(), d.props.speed, d.nif.pci_path.c_str(),
+              ptr_support_);
+  }
+}
+
+ncclResult_t Net::get_properties(int dev, ncclNetProperties_v6_t* props) {
+  if (dev < 0 || dev >= (int)devs_.size()) return ncclInvalidArgument;
+  *props = devs_[dev].props;
+  return ncclSuccess;
+}
+
+ncclResult_t Net::listen(int dev, void* handle, void** listen_comm) {
+  if (dev < 0 || dev >= (int)devs_.size()) return ncclInvalidArgument;
+  const sockaddr_storage& a = devs_[dev].nif.addr;
+  int af = a.ss_family;
+  int fd = socket(af, SOCK_STREAM | SOCK_CLOEXEC, 0);
+  if (fd < 0) return ncclSystemError;
+  int one = 1;
+  setsockopt(fd, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
+  socklen_t alen = af == AF_INET ? sizeof(sockaddr_in) : sizeof(sockaddr_in6);
+  sockaddr_storage bindaddr = a;  // port 0 → ephemeral
+  if (bind(fd, (sockaddr*)&bindaddr, alen) < 0 ||
+      ::listen(fd, Config::get().backlog) < 0 || set_nonblock2(fd) < 0) {
+    BNET_WARN("listen setup failed on %s: %s", devs_[dev].name_str.c_str(),
+              strerror(errno));
+    close(fd);
+    return ncclSystemError;
+  }
+  sockaddr_storage bound{};
+  socklen_t blen = sizeof(bound);
+  getsockname(fd, (sockaddr*)&bound, &blen);
+
+  auto* h = (ListenHandle*)handle;
+  memset(h, 0, sizeof(*h));
+  h->magic = kMagic;
+  h->family = (uint16_t)af;
+  if (af == AF_INET) {
+    auto* sin = (sockaddr_in*)&bound;
+    h->port = sin->sin_port;
+    memcpy(h->addr, &sin->sin_addr, 4);
+  } else {
+    auto* sin6 = (sockaddr_in6*)&bound;
+    h->port = sin6->sin6_port;
+    memcpy(h->addr, &sin6->sin6_addr, 16);
+  }
+
+  auto* l = new ListenComm();
+  l->fd = fd;
+  l->dev = dev;
+  *listen_comm = l;
+  return ncclSuccess;
+}
+
+static void fill_peer(const ListenHandle* h, sockaddr_storage* ss,
+                      socklen_t* len) {
+  memset(ss, 0, sizeof(*ss));
+  if (h->family == AF_INET) {
+    auto* sin = (sockaddr_in*)ss;
+    sin->sin_family = AF_INET;
+    sin->sin_port = h->port;
+    memcpy(&sin->sin_addr, h->addr, 4);
+    *len = sizeof(sockaddr_in);
+  } else {
+    auto* sin6 = (sockaddr_in6*)ss;
+    sin6->sin6_family = AF_INET6;
+    sin6->sin6_port = h->port;
+    memcpy(&sin6->sin6_addr, h->addr, 16);
+    *len = sizeof(sockaddr_in6);
+  }
+}
+
+ncclResult_t Net::connect(int dev, void* handle, void** send_comm) {
+  *send_comm = nullptr;
+  auto* h = (ListenHandle*)handle;
+  if (h->magic != kMagic) {
+    BNET_WARN("connect: bad handle magic 0x%x", h->magic);
+    return ncclInvalidArgument;
+  }
+  auto* t = (ConnectTask*)(uintptr_t)h->stage;
+  const Config& cfg = Config::get();
+  if (!t) {
+    t = new ConnectTask();
+    t->dev = dev;
+    fill_peer(h, &t->peer, &t->peer_len);
+    static std::atomic<uint64_t> ctr{0};
+    std::random_device rd;
+    t->conn_id = ((uint64_t)rd() << 32) ^ (uint64_t)rd() ^
+                 (ctr.fetch_add(1) << 1) ^ (uint64_t)getpid();
+    t->socks.resize(cfg.nstreams);
+    for (int i = 0; i < cfg.nstreams; i++) {
+      int fd = socket(t->peer.ss_family, SOCK_STREAM | SOCK_CLOEXEC, 0);
+      if (fd < 0) return ncclSystemError;
+      tune_socket2(fd);
+      // route via the chosen NIC
+      if (dev >= 0 && dev < (int)devs_.size()) {
+        sockaddr_storage local = devs_[dev].nif.addr;
+        socklen_t llen = local.ss_family == AF_INET ? sizeof(sockaddr_in)
+                                                    : sizeof(sockaddr_in6);
+        if (local.ss_family == t->peer.ss_family)
+          (void)bind(fd, (sockaddr*)&local, llen);
+      }
+      set_nonblock2(fd);
+      int rc = ::connect(fd, (sockaddr*)&t->peer, t->peer_len);
+      if (rc < 0 && errno != EINPROGRESS) {
+        BNET_WARN("connect() failed: %s", strerror(errno));
+        close(fd);
+        for (auto& p : t->socks)
+          if (p.fd >= 0 && p.fd != fd) close(p.fd);
+        delete t;
+        return ncclRemoteError;
+      }
+      t->socks[i].fd = fd;
+      t->socks[i].connected = (rc == 0);
+    }
+    h->stage = (uint64_t)(uintptr_t)t;
+  }
+
+  // progress all streams, nonblocking
+  bool all_done = true;
+  for (int i = 0; i < (int)t->socks.size(); i++) {
+    auto& p = t->socks[i];
+    if (!p.connected) {
+      struct pollfd pf = {p.fd, POLLOUT, 0};
+      if (poll(&pf, 1, 0) > 0 && (pf.revents & POLLOUT)) {
+        int err = 0;
+        socklen_t elen = sizeof(err);
+        getsockopt(p.fd, SOL_SOCKET, SO_ERROR, &err, &elen);
+        if (err != 0) {
+          BNET_WARN("connect to peer failed: %s", strerror(err));
+          for (auto& q : t->socks)
+            if (q.fd >= 0) close(q.fd);
+          delete t;
+          h->stage = 0;
+          return ncclRemoteError;
+        }
+        p.connected = true;
+      }
+    }
+    if (p.connected && p.hello_sent < sizeof(WireHello)) {
+      WireHello hello{kMagic, kWireVersion, t->conn_id, (uint16_t)i,
+                      (uint16_t)t->socks.size(), 0};
+      ssize_t w = ::send(p.fd, (char*)&hello + p.hello_sent,
+                         sizeof(hello) - p.hello_sent, MSG_NOSIGNAL);
+      if (w > 0) p.hello_sent += (uint32_t)w;
+      else if (w < 0 && errno != EAGAIN && errno != EWOULDBLOCK &&
+               errno != EINTR) {
+        BNET_WARN("hello write failed: %s", strerror(errno));
+        for (auto& q : t->socks)
+          if (q.fd >= 0) close(q.fd);
+        delete t;
+        h->stage = 0;
+        return ncclRemoteError;
+      }
+    }
+    if (!p.connected || p.hello_sent < sizeof(WireHello)) all_done = false;
+  }
+  if (!all_done) return ncclSuccess;  // *send_comm stays NULL → re-poll
+
+  auto* c = new SendComm();
+  c->dev = dev;
+  c->socks.reserve(t->socks.size());
+  for (auto& p : t->socks) {
+    auto* s = new TcpSock();
+    s->fd = p.fd;
+    s->is_recv = false;
+    s->scomm = c;
+    c->socks.push_back(s);
+  }
+  c->live_socks.store((int)c->socks.size());
+  for (auto* s : c->socks) Engine::get().register_sock(s);
+  delete t;
+  h->stage = 0;
+  *send_comm = c;
+  BNET_TRACE("send comm %p established (%d streams)", (void*)c,
+             (int)c->socks.size());
+  return ncclSuccess;
+}
+
+ncclResult_t Net::accept(void* listen_comm, void** recv_comm) {
+  *recv_comm = nullptr;
+  auto* l = (ListenComm*)listen_comm;
+
+  // 1. accept any pending connections
+  while (true) {
+    int fd = accept4(l->fd, nullptr, nullptr, SOCK_NONBLOCK | SOCK_CLOEXEC);
+    if (fd < 0) break;
+    tune_socket2(fd);
+    l->half.push_back({fd, {}, 0});
+  }
+
+  // 2. progress hello reads
+  for (auto it = l->half.begin(); it != l->half.end();) {
+    ssize_t n = read(it->fd, (char*)&it->hello + it->got,
+                     sizeof(WireHello) - it->got);
+    if (n > 0) it->got += (uint32_t)n;
+    if (it->got == sizeof(WireHello)) {
+      if (it->hello.magic != kMagic || it->hello.version != kWireVersion ||
+          it->hello.nstreams == 0 ||
+          it->hello.stream_id >= it->hello.nstreams) {
+        BNET_WARN("accept: bad hello (magic 0x%x)", it->hello.magic);
+        close(it->fd);
+      } else {
+        ListenComm::Group* g = nullptr;
+        for (auto& gg : l->groups)
+          if (gg.conn_id == it->hello.conn_id) g = &gg;
+        if (!g) {
+          l->groups.push_back({it->hello.conn_id, it->hello.nstreams,
+                               std::vector<int>(it->hello.nstreams, -1), 0});
+          g = &l->groups.back();
+        }
+        if (g->fds[it->hello.stream_id] == -1) {
+          g->fds[it->hello.stream_id] = it->fd;
+          g->have++;
+        } else {
+          close(it->fd);  // duplicate stream id — drop
+        }
+      }
+      it = l->half.erase(it);
+    } else if (n == 0 || (n < 0 && errno != EAGAIN && errno != EWOULDBLOCK &&
+                          errno != EINTR)) {
+      close(it->fd);
+      it = l->half.erase(it);
+    } else {
+      ++it;
+    }
+  }
+
+  // 3. a complete group → build the recv comm
+  for (auto it = l->groups.begin(); it != l->groups.end(); ++it) {
+    if (it->have != (int)it->nstreams) continue;
+    auto* c = new RecvComm();
+    c->dev = l->dev;
+    for (int fd : it->fds) {
+      auto* s = new TcpSock();
+      s->fd = fd;
+      s->is_recv = true;
+      s->rcomm = c;
+      c->socks.push_back(s);
+    }
+    c->live_socks.store((int)c->socks.size());
+    for (auto* s : c->socks) Engine::get().register_sock(s);
+    l->groups.erase(it);
+    *recv_comm = c;
+    BNET_TRACE("recv comm %p established (%d streams)", (void*)c,
+               (int)c->socks.size());
+    return ncclSuccess;
+  }
+  return ncclSuccess;
+}
+
+ncclResult_t Net::isend(void* send_comm, void* data, int size, int tag,
+                        void* mhandle, void** request) {
+  (void)tag;
+  auto* c = (SendComm*)send_comm;
+  if (c->error.load(std::memory_order_relaxed)) return ncclSystemError;
+  SendRequest* r = &c->reqs[c->seq_next % NCCL_NET_MAX_REQUESTS];
+  if (r->state.load(std::memory_order_acquire) != REQ_FREE) {
+    *request = nullptr;  // slot busy — NCCL retries
+    return ncclSuccess;
+  }
+  int ptr_type = (int)(uintptr_t)mhandle;
+  r->total = (uint32_t)size;
+  r->chunk = pick_chunk_size((uint32_t)size, Config::get().min_chunk,
+                             Config::get().max_chunk,
+                             (int)c->socks.size());
+  r->cursor.store(0, std::memory_order_relaxed);
+  r->sent.store(0, std::memory_order_relaxed);
+  r->hdr_claimed.store(false, std::memory_order_relaxed);
+  r->hdr_sent.store(false, std::memory_order_relaxed);
+  r->comm = c;
+  if (ptr_type == NCCL_PTR_CUDA && size > 0) {
+    if (!c->stage_pool) {
+      c->stage_pool = stage_pool_create();
+      if (!c->stage_pool) return ncclInternalError;
+    }
+    if (!stage_send_begin(c->stage_pool, r, data, (uint32_t)size)) {
+      *request = nullptr;  // pool exhausted — retry later
+      return ncclSuccess;
+    }
+    r->avail.store(0, std::memory_order_relaxed);
+  } else {
+    r->src = (const char*)data;
+    r->avail.store((uint32_t)size, std::memory_order_relaxed);
+  }
+  r->seq.store(c->seq_next, std::memory_order_relaxed);
+  r->state.store(REQ_ACTIVE, std::memory_order_release);
+  c->seq_next++;
+  c->stats.isend_count.fetch_add(1, std::memory_order_relaxed);
+  Engine::get().kick_comm(c);
+  *request = tag_send(r);
+  return ncclSuccess;
+}
+
+ncclResult_t Net::irecv(void* recv_comm, int n, void** data, int* sizes,
+                        int* tags, void** mhandles, void** request) {
+  (void)tags;
+  auto* c = (RecvComm*)recv_comm;
+  if (c->error.load(std::memory_order_relaxed)) return ncclSystemError;
+  if (n != 1) {
+    BNET_WARN("irecv: grouped recv n=%d unsupported (maxRecvs=1)", n);
+    return ncclInternalError;
+  }
+  RecvRequest* r = &c->reqs[c->post_next % NCCL_NET_MAX_REQUESTS];
+  if (r->state.load(std::memory_order_acquire) != REQ_FREE) {
+    *request = nullptr;
+    return ncclSuccess;
+  }
+  int ptr_type = mhandles ? (int)(uintptr_t)mhandles[0] : NCCL_PTR_HOST;
+  r->dst = (char*)data[0];
+  r->capacity = (uint32_t)sizes[0];
+  r->total.store(-1, std::memory_order_relaxed);
+  r->received.store(0, std::memory_order_relaxed);
+  r->gpu_done.store(false, std::memory_order_relaxed);
+  r->comm = c;
+  if (ptr_type == NCCL_PTR_CUDA) {
+    if (!c->stage_pool) {
+      c->stage_pool = stage_pool_create();
+      if (!c->stage_pool) return ncclInternalError;
+    }
+    if (!stage_recv_begin(c->stage_pool, r, r->dst, r->capacity)) {
+      *request = nullptr;
+      return ncclSuccess;
+    }
+  } else {
+    r->stage = nullptr;
+  }
+  r->seq.store(c->post_next, std::memory_order_relaxed);
+  r->state.store(REQ_ACTIVE, std::memory_order_release);
+  c->post_next++;
+  c->stats.irecv_count.fetch_add(1, std::memory_order_relaxed);
+  Engine::get().kick_comm(c);  // wake sockets parked on this seq
+  *request = tag_recv(r);
+  return ncclSuccess;
+}
+
+ncclResult_t Net::iflush(void* recv_comm, int n, void** data, int* sizes,
+                         void** mhandles, void** request) {
+  // Staged recvs complete only after their H2D copies drain, so received
+  // data is already visible to the GPU — nothing to flush.
+  (void)recv_comm, (void)n, (void)data, (void)sizes, (void)mhandles;
+  *request = nullptr;
+  return ncclSuccess;
+}
+
+ncclResult_t Net::test(void* request, int* done, int* sizes) {
+  *done = 0;
+  uintptr_t v = (uintptr_t)request;
+  if (v & 1) {
+    auto* r = (SendRequest*)(v & ~(uintptr_t)7);
+    SendComm* c = r->comm;
+    if (c->error.load(std::memory_order_relaxed)) {
+      BNET_WARN("test(send): comm error %d", c->error.load());
+      return ncclSystemError;
+    }
+    if (c->stage_pool && stage_pending(c->stage_pool))
+      stage_poll(c->stage_pool);
+    if (r->complete()) {
+      *done = 1;
+      if (sizes) sizes[0] = (int)r->total;
+      if (r->stage) stage_release(c->stage_pool, r);
+      r->state.store(REQ_FREE, std::memory_order_release);
+    }
+    return ncclSuccess;
+  }
+  if (v & 2) {
+    auto* r = (RecvRequest*)(v & ~(uintptr_t)7);
+    RecvComm* c = r->comm;
+    if (c->error.load(std::memory_order_relaxed)) {
+      BNET_WARN("test(recv): comm error %d", c->error.load());
+      return ncclSystemError;
+    }
+    if (r->socket_complete() && (!r->stage || stage_recv_done(r))) {
+      *done = 1;
+      if (sizes) sizes[0] = (int)r->total.load(std::memory_order_acquire);
+      if (r->stage) stage_release(c->stage_pool, r);
+      r->state.store(REQ_FREE, std::memory_order_release);
+    }
+    return ncclSuccess;
+  }
+  return ncclInvalidArgument;
+}
+
+ncclResult_t Net::close_send(void* send_comm) {
+  auto* c = (SendComm*)send_comm;
+  for (auto* s : c->socks) {
+    Engine::get().unregister_sock_sync(s);
+    delete s;
+  }
+  if (c->stage_pool) stage_pool_destroy(c->stage_pool);
+  delete c;
+  return ncclSuccess;
+}
+
+ncclResult_t Net::close_recv(void* recv_comm) {
+  auto* c = (RecvComm*)recv_comm;
+  for (auto* s : c->socks) {
+    Engine::get().unregister_sock_sync(s);
+    delete s;
+  }
+  if (c->stage_pool) stage_pool_destroy(c->stage_pool);
+  delete c;
+  return ncclSuccess;
+}
+
+ncclResult_t Net::close_listen(void* listen_comm) {
+  auto* l = (ListenComm*)listen_comm;
+  if (l->fd >= 0) close(l->fd);
+  for (auto& h : l->half) close(h.fd);
+  for (auto& g : l->groups)
+    for (int fd : g.fds)
+      if (fd >= 0) close(fd);
+  delete l;
+  return ncclSuccess;
+}
+
+}  // namespace baguanet

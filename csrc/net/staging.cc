@@ -1,0 +1,271 @@
+// staging.cc — pinned-ring staging implementation (see staging.h).
+
+#include "staging.h"
+
+#include <hip/hip_runtime.h>
+
+#include <algorithm>
+#include <mutex>
+#include <vector>
+
+#include "baguanet/config.h"
+#include "baguanet/log.h"
+#include "transport.h"
+
+namespace baguanet {
+
+// pack_kernels.hip — vectorized CDNA4 copy kernels (BNET_STAGE_KERNEL=1).
+void launch_copy_kernel(void* dst, const void* src, size_t bytes,
+                        hipStream_t stream);
+
+#define HIP_WARN(call)                                             \
+  do {                                                             \
+    hipError_t e_ = (call);                                        \
+    if (e_ != hipSuccess)                                          \
+      BNET_WARN("%s failed: %s", #call, hipGetErrorString(e_));    \
+  } while (0)
+
+struct StageAlloc {
+  char* host = nullptr;
+  size_t pool_off = 0;
+  uint32_t size = 0;
+  // send pipeline
+  const char* gpu_src = nullptr;
+  uint32_t total = 0;
+  uint32_t copy_chunk = 0;
+  std::vector<hipEvent_t> events;
+  uint32_t events_done = 0;
+  SendRequest* sreq = nullptr;
+  // recv pipeline
+  char* gpu_dst = nullptr;
+  hipEvent_t done_ev = nullptr;
+};
+
+class StagePool {
+ public:
+  char* base = nullptr;
+  size_t size = 0;
+  int device = 0;
+  hipStream_t d2h = nullptr, h2d = nullptr;
+  std::mutex mu;
+  struct Range {
+    size_t off, len;
+  };
+  std::vector<Range> free_list;
+  std::vector<hipEvent_t> ev_cache;
+  std::vector<StageAlloc*> inflight;  // send stagings with copies pending
+  std::atomic<int> pending{0};
+
+  hipEvent_t get_event() {
+    if (!ev_cache.empty()) {
+      hipEvent_t e = ev_cache.back();
+      ev_cache.pop_back();
+      return e;
+    }
+    hipEvent_t e = nullptr;
+    HIP_WARN(hipEventCreateWithFlags(&e, hipEventDisableTiming));
+    return e;
+  }
+  void put_event(hipEvent_t e) { ev_cache.push_back(e); }
+
+  char* alloc(uint32_t sz, size_t* off_out) {
+    size_t need = (sz + 255) & ~size_t(255);
+    if (need == 0) need = 256;
+    for (auto it = free_list.begin(); it != free_list.end(); ++it) {
+      if (it->len >= need) {
+        size_t off = it->off;
+        it->off += need;
+        it->len -= need;
+        if (it->len == 0) free_list.erase(it);
+        *off_out = off;
+        return base + off;
+      }
+    }
+    return nullptr;
+  }
+  void free(size_t off, uint32_t sz) {
+    size_t need = (sz + 255) & ~size_t(255);
+    if (need == 0) need = 256;
+    Range r{off, need};
+    auto it = std::lower_bound(
+        free_list.begin(), free_list.end(), r,
+        [](const Range& a, const Range& b) { return a.off < b.off; });
+    it = free_list.insert(it, r);
+    // coalesce with neighbors
+    if (it + 1 != free_list.end() && it->off + it->len == (it + 1)->off) {
+      it->len += (it + 1)->len;
+      free_list.erase(it + 1);
+    }
+    if (it != free_list.begin() && (it - 1)->off + (it - 1)->len == it->off) {
+      (it - 1)->len += it->len;
+      free_list.erase(it);
+    }
+  }
+};
+
+bool staging_available() {
+  if (!Config::get().cuda_ptr) return false;
+  static int avail = -1;
+  if (avail < 0) {
+    int n = 0;
+    avail = (hipGetDeviceCount(&n) == hipSuccess && n > 0) ? 1 : 0;
+  }
+  return avail == 1;
+}
+
+StagePool* stage_pool_create() {
+  auto* p = new StagePool();
+  p->size = Config::get().stage_pool;
+  hipError_t e = hipHostMalloc((void**)&p->base, p->size, hipHostMallocDefault);
+  if (e != hipSuccess) {
+    BNET_WARN("hipHostMalloc(%zu) failed: %s", p->size, hipGetErrorString(e));
+    delete p;
+    return nullptr;
+  }
+  HIP_WARN(hipGetDevice(&p->device));
+  HIP_WARN(hipStreamCreateWithFlags(&p->d2h, hipStreamNonBlocking));
+  HIP_WARN(hipStreamCreateWithFlags(&p->h2d, hipStreamNonBlocking));
+  p->free_list.push_back({0, p->size});
+  return p;
+}
+
+void stage_pool_destroy(StagePool* p) {
+  if (!p) return;
+  if (p->d2h) (void)hipStreamSynchronize(p->d2h);
+  if (p->h2d) (void)hipStreamSynchronize(p->h2d);
+  for (auto e : p->ev_cache) (void)hipEventDestroy(e);
+  if (p->d2h) (void)hipStreamDestroy(p->d2h);
+  if (p->h2d) (void)hipStreamDestroy(p->h2d);
+  if (p->base) (void)hipHostFree(p->base);
+  delete p;
+}
+
+static void issue_copy(void* dst, const void* src, size_t n, hipMemcpyKind k,
+                       hipStream_t s) {
+  if (Config::get().stage_kernel) {
+    launch_copy_kernel(dst, src, n, s);
+  } else {
+    HIP_WARN(hipMemcpyAsync(dst, src, n, k, s));
+  }
+}
+
+bool stage_send_begin(StagePool* p, SendRequest* req, const void* src,
+                      uint32_t total) {
+  std::lock_guard<std::mutex> lk(p->mu);
+  auto* a = new StageAlloc();
+  if (!(a->host = p->alloc(std::max(total, 1u), &a->pool_off))) {
+    delete a;
+    return false;
+  }
+  a->size = std::max(total, 1u);
+  a->gpu_src = (const char*)src;
+  a->total = total;
+  a->copy_chunk = Config::get().stage_chunk;
+  a->sreq = req;
+  req->src = a->host;
+  req->stage = a;
+  for (uint32_t off = 0; off < total; off += a->copy_chunk) {
+    uint32_t n = std::min(a->copy_chunk, total - off);
+    issue_copy(a->host + off, a->gpu_src + off, n, hipMemcpyDeviceToHost,
+               p->d2h);
+    hipEvent_t ev = p->get_event();
+    HIP_WARN(hipEventRecord(ev, p->d2h));
+    a->events.push_back(ev);
+  }
+  if (total > 0) {
+    p->inflight.push_back(a);
+    p->pending.fetch_add(1, std::memory_order_release);
+  }
+  return true;
+}
+
+bool stage_poll(StagePool* p) {
+  if (p->pending.load(std::memory_order_acquire) == 0) return false;
+  std::lock_guard<std::mutex> lk(p->mu);
+  bool advanced = false;
+  for (auto it = p->inflight.begin(); it != p->inflight.end();) {
+    StageAlloc* a = *it;
+    while (a->events_done < a->events.size() &&
+           hipEventQuery(a->events[a->events_done]) == hipSuccess) {
+      a->events_done++;
+      advanced = true;
+      uint32_t avail =
+          std::min<uint64_t>((uint64_t)a->events_done * a->copy_chunk,
+                             a->total);
+      a->sreq->avail.store(avail, std::memory_order_release);
+    }
+    if (a->events_done == a->events.size()) {
+      for (auto e : a->events) p->put_event(e);
+      a->events.clear();
+      it = p->inflight.erase(it);
+      p->pending.fetch_sub(1, std::memory_order_release);
+    } else {
+      ++it;
+    }
+  }
+  return advanced;
+}
+
+bool stage_pending(StagePool* p) {
+  return p->pending.load(std::memory_order_acquire) > 0;
+}
+
+bool stage_recv_begin(StagePool* p, RecvRequest* req, void* dst,
+                      uint32_t capacity) {
+  std::lock_guard<std::mutex> lk(p->mu);
+  auto* a = new StageAlloc();
+  if (!(a->host = p->alloc(std::max(capacity, 1u), &a->pool_off))) {
+    delete a;
+    return false;
+  }
+  a->size = std::max(capacity, 1u);
+  a->gpu_dst = (char*)dst;
+  req->stage = a;
+  return true;
+}
+
+char* stage_recv_base(RecvRequest* req) {
+  return req->stage ? ((StageAlloc*)req->stage)->host : req->dst;
+}
+
+void stage_recv_chunk(RecvRequest* req, uint32_t offset, uint32_t len,
+                      bool last) {
+  StageAlloc* a = (StageAlloc*)req->stage;
+  StagePool* p = req->comm->stage_pool;
+  std::lock_guard<std::mutex> lk(p->mu);
+  if (len)
+    issue_copy(a->gpu_dst + offset, a->host + offset, len,
+               hipMemcpyHostToDevice, p->h2d);
+  if (last) {
+    a->done_ev = p->get_event();
+    HIP_WARN(hipEventRecord(a->done_ev, p->h2d));
+  }
+}
+
+bool stage_recv_done(RecvRequest* req) {
+  StageAlloc* a = (StageAlloc*)req->stage;
+  if (!a) return true;
+  if (!a->done_ev) return false;  // last chunk not yet landed
+  return hipEventQuery(a->done_ev) == hipSuccess;
+}
+
+void stage_release(StagePool* p, SendRequest* req) {
+  StageAlloc* a = (StageAlloc*)req->stage;
+  if (!a) return;
+  std::lock_guard<std::mutex> lk(p->mu);
+  p->free(a->pool_off, a->size);
+  req->stage = nullptr;
+  delete a;
+}
+
+void stage_release(StagePool* p, RecvRequest* req) {
+  StageAlloc* a = (StageAlloc*)req->stage;
+  if (!a) return;
+  std::lock_guard<std::mutex> lk(p->mu);
+  if (a->done_ev) p->put_event(a->done_ev);
+  p->free(a->pool_off, a->size);
+  req->stage = nullptr;
+  delete a;
+}
+
+}  // namespace baguanet

@@ -1,0 +1,156 @@
+"""CPU loopback tests of the real ncclNetPlugin_v6 vtable (BASELINE config 1:
+"ncclNet connect/isend/irecv over loopback TCP, 2 CPU ranks, no GPU").
+
+These cover the wire protocol end to end — the reference's biggest test gap
+(it never tested its own wire protocol).
+"""
+
+import ctypes as C
+import os
+import random
+import time
+
+import pytest
+
+
+def establish(plugin, dev=0, timeout=10.0):
+    """Nonblocking connect/accept pumped from one thread until both sides
+    are up (exactly how the RCCL proxy drives the plugin)."""
+    handle, lcomm = plugin.listen(dev)
+    scomm = rcomm = None
+    t0 = time.monotonic()
+    while scomm is None or rcomm is None:
+        if scomm is None:
+            scomm = plugin.connect(dev, handle)
+        if rcomm is None:
+            rcomm = plugin.accept(lcomm)
+        if time.monotonic() - t0 > timeout:
+            raise TimeoutError("connection setup did not complete")
+    return lcomm, scomm, rcomm
+
+
+@pytest.fixture(scope="module")
+def conn(plugin):
+    lcomm, scomm, rcomm = establish(plugin)
+    yield plugin, scomm, rcomm
+    plugin.close_send(scomm)
+    plugin.close_recv(rcomm)
+    plugin.close_listen(lcomm)
+
+
+def test_init_devices_properties(plugin):
+    assert plugin.name == "BaguaNetAMD"
+    n = plugin.ndev()
+    assert n >= 1
+    props = plugin.properties(0)
+    assert props["name"] == "lo"
+    assert props["ptrSupport"] & 0x1  # NCCL_PTR_HOST
+    assert props["maxComms"] > 0
+    assert props["maxRecvs"] == 1
+    assert props["speed"] > 0
+
+
+def xfer(plugin, scomm, rcomm, payload: bytes, recv_pad: int = 0):
+    """One message send->recv; returns received bytes."""
+    size = len(payload)
+    sbuf = C.create_string_buffer(payload, max(size, 1))
+    rbuf = C.create_string_buffer(size + recv_pad + 1)
+    smh = plugin.reg_mr(scomm, sbuf, size)
+    rmh = plugin.reg_mr(rcomm, rbuf, size + recv_pad)
+
+    rreq = None
+    sreq = None
+    t0 = time.monotonic()
+    while rreq is None:
+        rreq = plugin.irecv(rcomm, rbuf, size + recv_pad, rmh)
+        assert time.monotonic() - t0 < 10
+    while sreq is None:
+        sreq = plugin.isend(scomm, sbuf, size, smh)
+        assert time.monotonic() - t0 < 10
+
+    ssize = plugin.wait(sreq)
+    rsize = plugin.wait(rreq)
+    assert ssize == size
+    assert rsize == size
+    return rbuf.raw[:size]
+
+
+@pytest.mark.parametrize(
+    "size",
+    [0, 1, 7, 100, 4096, 8192, 65536, 1 << 20, (1 << 22) + 13],
+)
+def test_roundtrip_sizes(conn, size):
+    plugin, scomm, rcomm = conn
+    rng = random.Random(size)
+    payload = bytes(rng.getrandbits(8) for _ in range(min(size, 1 << 16)))
+    if size > len(payload):  # big sizes: repeat a random block (cheap)
+        payload = (payload * (size // max(len(payload), 1) + 1))[:size]
+    got = xfer(plugin, scomm, rcomm, payload)
+    assert got == payload
+
+
+def test_recv_larger_than_send(conn):
+    plugin, scomm, rcomm = conn
+    payload = os.urandom(10000)
+    got = xfer(plugin, scomm, rcomm, payload, recv_pad=5000)
+    assert got == payload
+
+
+def test_many_outstanding_messages(conn):
+    plugin, scomm, rcomm = conn
+    n_msgs = 100
+    msgs = [os.urandom(1000 + i * 37) for i in range(n_msgs)]
+    sbufs = [C.create_string_buffer(m, len(m)) for m in msgs]
+    rbufs = [C.create_string_buffer(len(m) + 1) for m in msgs]
+    smh = plugin.reg_mr(scomm, None, 0)
+    rmh = plugin.reg_mr(rcomm, None, 0)
+
+    sreqs, rreqs = {}, {}
+    si = ri = 0
+    t0 = time.monotonic()
+    # Post sends and recvs as slots free up, completing as we go — mirrors
+    # the proxy's pipelined outstanding-request pattern.
+    while len(sreqs) < n_msgs or len(rreqs) < n_msgs or any(
+        v is not None for v in list(sreqs.values()) + list(rreqs.values())
+    ):
+        assert time.monotonic() - t0 < 60
+        if ri < n_msgs:
+            r = plugin.irecv(rcomm, rbufs[ri], len(msgs[ri]), rmh)
+            if r is not None:
+                rreqs[ri] = r
+                ri += 1
+        if si < n_msgs:
+            s = plugin.isend(scomm, sbufs[si], len(msgs[si]), smh)
+            if s is not None:
+                sreqs[si] = s
+                si += 1
+        for d, reqs in ((sreqs, sreqs), (rreqs, rreqs)):
+            for k, req in list(reqs.items()):
+                if req is None:
+                    continue
+                done, size = plugin.test(req)
+                if done:
+                    assert size == len(msgs[k])
+                    reqs[k] = None
+    for i, m in enumerate(msgs):
+        assert rbufs[i].raw[: len(m)] == m, f"message {i} corrupted"
+
+
+def test_multiple_comms(plugin):
+    conns = [establish(plugin) for _ in range(4)]
+    for i, (lc, sc, rc) in enumerate(conns):
+        payload = bytes([i * 17 % 256]) * (50000 + i)
+        got = xfer(plugin, sc, rc, payload)
+        assert got == payload
+    for lc, sc, rc in conns:
+        plugin.close_send(sc)
+        plugin.close_recv(rc)
+        plugin.close_listen(lc)
+
+
+def test_zero_then_data(conn):
+    plugin, scomm, rcomm = conn
+    assert xfer(plugin, scomm, rcomm, b"") == b""
+    payload = os.urandom(123457)
+    assert xfer(plugin, scomm, rcomm, payload) == payload
+    assert xfer(plugin, scomm, rcomm, b"") == b""
