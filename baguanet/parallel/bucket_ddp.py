@@ -1,0 +1,146 @@
+"""BucketedDDP — data-parallel gradient all-reduce with bucket fusion and
+comm/compute overlap, over torch.distributed (RCCL on ROCm, gloo on CPU).
+
+This is the layer *above* the transport in the reference's benchmark stack
+(Bagua/PyTorch-DDP gradient-bucket allreduce -> NCCL ring ->
+bagua-net).  MI355X-first choices:
+
+* Gradients live directly in fused flat bucket buffers (``p.grad`` is a
+  view), so there is no pack step in the hot path and each bucket is ONE
+  contiguous all-reduce — sized for xGMI's per-link-bound ring (7 p2p links
+  x ~153 GB/s): default 50 MiB buckets keep per-rank link transfers large
+  enough to amortize ring latency while still overlapping with backward.
+* All-reduce launches per bucket from post-accumulate-grad hooks, in
+  backward order, on the communicator's own stream — overlap comes from
+  RCCL running on a separate stream/queue, not from CPU threads.
+"""
+
+from __future__ import annotations
+
+import contextlib
+from typing import List, Optional
+
+import torch
+import torch.distributed as dist
+from torch import nn
+
+
+class _Bucket:
+    def __init__(self, params: List[nn.Parameter], device, dtype):
+        self.params = params
+        self.numels = [p.numel() for p in params]
+        self.total = sum(self.numels)
+        self.flat = torch.zeros(self.total, device=device, dtype=dtype)
+        self.pending = len(params)
+        self.work: Optional[dist.Work] = None
+        # carve grad views
+        off = 0
+        for p in params:
+            n = p.numel()
+            p.grad = self.flat[off : off + n].view_as(p)
+            off += n
+
+
+class BucketedDDP(nn.Module):
+    """Wrap a module for data-parallel training.
+
+    Usage::
+
+        model = BucketedDDP(model)           # after dist.init_process_group
+        out = model(x); loss.backward()
+        model.finish_backward()              # wait for async all-reduces
+        optimizer.step(); model.zero_grad()  # NOT optimizer.zero_grad(
+                                             # set_to_none=True) — grads are
+                                             # bucket views
+
+    Matches DDP numerics: gradients are averaged over ranks.
+    """
+
+    def __init__(
+        self,
+        module: nn.Module,
+        bucket_cap_mb: float = 50.0,
+        process_group=None,
+        broadcast_params: bool = True,
+    ):
+        super().__init__()
+        self.module = module
+        self.group = process_group
+        self.world = dist.get_world_size(process_group) if dist.is_initialized() else 1
+        self._sync_enabled = True
+        if broadcast_params and self.world > 1:
+            with torch.no_grad():
+                for p in module.parameters():
+                    dist.broadcast(p.data, src=0, group=self.group)
+            for b in module.buffers():
+                dist.broadcast(b.data, src=0, group=self.group)
+
+        # Build buckets in REVERSE parameter order: backward produces
+        # gradients roughly from the last layer backwards, so reverse-order
+        # buckets fill (and launch) earliest during backward.
+        params = [p for p in module.parameters() if p.requires_grad]
+        cap = int(bucket_cap_mb * 1024 * 1024)
+        self.buckets: List[_Bucket] = []
+        cur: List[nn.Parameter] = []
+        cur_bytes = 0
+        for p in reversed(params):
+            sz = p.numel() * p.element_size()
+            if cur and cur_bytes + sz > cap:
+                self.buckets.append(self._make_bucket(cur))
+                cur, cur_bytes = [], 0
+            cur.append(p)
+            cur_bytes += sz
+        if cur:
+            self.buckets.append(self._make_bucket(cur))
+
+        self._param_bucket = {}
+        for b in self.buckets:
+            for p in b.params:
+                self._param_bucket[id(p)] = b
+        self._hooks = [
+            p.register_post_accumulate_grad_hook(self._grad_ready)
+            for b in self.buckets
+            for p in b.params
+        ]
+
+    @staticmethod
+    def _make_bucket(params: List[nn.Parameter]) -> _Bucket:
+        p0 = params[0]
+        return _Bucket(list(params), p0.device, p0.dtype)
+
+    def _grad_ready(self, param: nn.Parameter) -> None:
+        b = self._param_bucket[id(param)]
+        b.pending -= 1
+        if b.pending == 0:
+            b.pending = len(b.params)  # re-arm for the next backward
+            if self._sync_enabled and self.world > 1:
+                # pre-divide + SUM == AVG, works on both gloo and nccl
+                b.flat.div_(self.world)
+                b.work = dist.all_reduce(
+                    b.flat, op=dist.ReduceOp.SUM, group=self.group,
+                    async_op=True,
+                )
+
+    def forward(self, *args, **kwargs):
+        return self.module(*args, **kwargs)
+
+    def finish_backward(self) -> None:
+        """Wait for all in-flight bucket all-reduces."""
+        for b in self.buckets:
+            if b.work is not None:
+                b.work.wait()
+                b.work = None
+
+    def zero_grad(self) -> None:
+        for b in self.buckets:
+            b.flat.zero_()
+
+    @contextlib.contextmanager
+    def no_sync(self):
+        """Skip gradient synchronization (for gradient accumulation)."""
+        prev = self._sync_enabled
+        self._sync_enabled = False
+        try:
+            yield
+        finally:
+            self._sync_enabled = prev

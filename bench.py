@@ -1,0 +1,154 @@
+#!/usr/bin/env python3
+"""bench.py — flagship benchmark: VGG16 synthetic data-parallel training
+(the reference's headline end-to-end benchmark: Bagua synthetic_benchmark.py
+VGG16 img/sec).
+
+Per-GPU batch 32, fp32 (the reference benchmark's dtype), synthetic data,
+random-init weights, BucketedDDP gradient all-reduce over RCCL (xGMI
+intra-node; the baguanet plugin carries any inter-node TCP leg and is
+loaded via NCCL_NET_PLUGIN).
+
+Run:  python bench.py [--gpus N] [--steps K] [--warmup W]
+Multi-GPU (driver): torch.distributed.run --nproc-per-node N bench.py ...
+Prints ONE JSON line from rank 0.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+# Reference numbers: VGG16 synthetic on 4x8xV100 100GbE,
+# 126.5 img/sec/GPU with bagua-net (4046.6 total at 32 GPUs).
+BASELINE_PER_GPU = 126.5
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=30)
+    ap.add_argument("--warmup", type=int, default=10)
+    ap.add_argument("--batch", type=int, default=32, help="per-GPU batch")
+    ap.add_argument("--model", default="vgg16",
+                    choices=["vgg16", "resnet50"])
+    ap.add_argument("--dtype", default="fp32", choices=["fp32", "bf16"])
+    ap.add_argument("--force-net", action="store_true",
+                    help="route even intra-node traffic through the plugin")
+    args = ap.parse_args()
+
+    # plugin env must be set before the first collective
+    from baguanet.plugin import rccl_env
+
+    for k, v in rccl_env(env={}, force_net=args.force_net).items():
+        if k == "LD_LIBRARY_PATH":
+            os.environ[k] = f"{v}:{os.environ.get(k, '')}".rstrip(":")
+        else:
+            os.environ.setdefault(k, v)
+
+    import torch
+    import torch.distributed as dist
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    n_gpus = world if world > 1 else args.gpus
+
+    use_cuda = torch.cuda.is_available()
+    device = torch.device(f"cuda:{local_rank}" if use_cuda else "cpu")
+    if use_cuda:
+        torch.cuda.set_device(device)
+    if world > 1:
+        dist.init_process_group(
+            "nccl" if use_cuda else "gloo", rank=rank, world_size=world
+        )
+
+    from baguanet.models import resnet50, vgg16
+    from baguanet.parallel import BucketedDDP
+
+    torch.manual_seed(42 + rank)
+    dtype = torch.float32 if args.dtype == "fp32" else torch.bfloat16
+    model = (vgg16() if args.model == "vgg16" else resnet50()).to(device)
+    if dtype != torch.float32:
+        model = model.to(dtype)
+    model = BucketedDDP(model, bucket_cap_mb=50.0)
+    opt = torch.optim.SGD(model.module.parameters(), lr=0.01, momentum=0.9)
+
+    x = torch.randn(args.batch, 3, 224, 224, device=device, dtype=dtype)
+    y = torch.randint(0, 1000, (args.batch,), device=device)
+    loss_fn = torch.nn.CrossEntropyLoss()
+
+    def step():
+        model.zero_grad()
+        out = model(x)
+        loss = loss_fn(out.float(), y)
+        loss.backward()
+        model.finish_backward()
+        opt.step()
+
+    for _ in range(args.warmup):
+        step()
+
+    if world > 1:
+        dist.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    if use_cuda:
+        torch.cuda.synchronize()
+    if world > 1:
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    if world > 1:
+        t = torch.tensor([elapsed], device=device if use_cuda else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    total_img_per_sec = args.batch * n_gpus * args.steps / elapsed
+
+    if rank == 0:
+        out = {
+            "metric": "VGG16 synthetic training img/sec (total)"
+            if args.model == "vgg16"
+            else f"{args.model} synthetic training img/sec (total)",
+            "value": round(total_img_per_sec, 1),
+            "unit": "img/sec",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": round(
+                total_img_per_sec / (BASELINE_PER_GPU * n_gpus), 3
+            ),
+            "dtype": args.dtype,
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": args.batch * n_gpus,
+                "seq_len": None,
+                "image": "3x224x224",
+                "parallelism": f"dp{n_gpus}",
+                "baseline": "bagua-net VGG16 126.5 img/sec/GPU "
+                "(4x8xV100 100GbE, BASELINE.md) scaled to n_gpus",
+            },
+        }
+        print(json.dumps(out))
+
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
