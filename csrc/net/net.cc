@@ -18,6 +18,7 @@
 
 #include "baguanet/log.h"
 #include "staging.h"
+#include "telemetry.h"
 #include "transport.h"
 
 namespace baguanet {
@@ -243,6 +244,7 @@ ncclResult_t Net::connect(int dev, void* handle, void** send_comm) {
   delete t;
   h->stage = 0;
   *send_comm = c;
+  Telemetry::get().send_comms.fetch_add(1, std::memory_order_relaxed);
   BNET_TRACE("send comm %p established (%d streams)", (void*)c,
              (int)c->socks.size());
   return ncclSuccess;
@@ -313,6 +315,7 @@ ncclResult_t Net::accept(void* listen_comm, void** recv_comm) {
     for (auto* s : c->socks) Engine::get().register_sock(s);
     l->groups.erase(it);
     *recv_comm = c;
+    Telemetry::get().recv_comms.fetch_add(1, std::memory_order_relaxed);
     BNET_TRACE("recv comm %p established (%d streams)", (void*)c,
                (int)c->socks.size());
     return ncclSuccess;
@@ -358,7 +361,13 @@ ncclResult_t Net::isend(void* send_comm, void* data, int size, int tag,
   r->state.store(REQ_ACTIVE, std::memory_order_release);
   c->seq_next++;
   c->stats.isend_count.fetch_add(1, std::memory_order_relaxed);
-  Engine::get().kick_comm(c);
+  auto& T = Telemetry::get();
+  T.isend_count.fetch_add(1, std::memory_order_relaxed);
+  T.hist_add(T.isend_hist, (uint64_t)size);
+  r->span_slot = T.span_begin(0, (uint64_t)(uintptr_t)c, r->seq.load(),
+                              (uint32_t)size);
+  int nchunks = r->total ? (int)((r->total + r->chunk - 1) / r->chunk) : 1;
+  Engine::get().kick_comm(c, nchunks);
   *request = tag_send(r);
   return ncclSuccess;
 }
@@ -400,6 +409,11 @@ ncclResult_t Net::irecv(void* recv_comm, int n, void** data, int* sizes,
   r->state.store(REQ_ACTIVE, std::memory_order_release);
   c->post_next++;
   c->stats.irecv_count.fetch_add(1, std::memory_order_relaxed);
+  auto& T = Telemetry::get();
+  T.irecv_count.fetch_add(1, std::memory_order_relaxed);
+  T.hist_add(T.irecv_hist, (uint64_t)sizes[0]);
+  r->span_slot = T.span_begin(1, (uint64_t)(uintptr_t)c, r->seq.load(),
+                              (uint32_t)sizes[0]);
   Engine::get().kick_comm(c);  // wake sockets parked on this seq
   *request = tag_recv(r);
   return ncclSuccess;
@@ -429,6 +443,9 @@ ncclResult_t Net::test(void* request, int* done, int* sizes) {
     if (r->complete()) {
       *done = 1;
       if (sizes) sizes[0] = (int)r->total;
+      auto& T = Telemetry::get();
+      T.bytes_sent.fetch_add(r->total, std::memory_order_relaxed);
+      T.span_end(r->span_slot);
       if (r->stage) stage_release(c->stage_pool, r);
       r->state.store(REQ_FREE, std::memory_order_release);
     }
@@ -444,6 +461,10 @@ ncclResult_t Net::test(void* request, int* done, int* sizes) {
     if (r->socket_complete() && (!r->stage || stage_recv_done(r))) {
       *done = 1;
       if (sizes) sizes[0] = (int)r->total.load(std::memory_order_acquire);
+      auto& T = Telemetry::get();
+      T.bytes_recv.fetch_add((uint64_t)r->total.load(),
+                             std::memory_order_relaxed);
+      T.span_end(r->span_slot);
       if (r->stage) stage_release(c->stage_pool, r);
       r->state.store(REQ_FREE, std::memory_order_release);
     }

@@ -10,6 +10,7 @@
 
 #include "baguanet/config.h"
 #include "baguanet/log.h"
+#include "telemetry.h"
 #include "transport.h"
 
 namespace baguanet {
@@ -168,6 +169,7 @@ bool stage_send_begin(StagePool* p, SendRequest* req, const void* src,
     uint32_t n = std::min(a->copy_chunk, total - off);
     issue_copy(a->host + off, a->gpu_src + off, n, hipMemcpyDeviceToHost,
                p->d2h);
+    Telemetry::get().staged_d2h_bytes.fetch_add(n, std::memory_order_relaxed);
     hipEvent_t ev = p->get_event();
     HIP_WARN(hipEventRecord(ev, p->d2h));
     a->events.push_back(ev);
@@ -233,9 +235,12 @@ void stage_recv_chunk(RecvRequest* req, uint32_t offset, uint32_t len,
   StageAlloc* a = (StageAlloc*)req->stage;
   StagePool* p = req->comm->stage_pool;
   std::lock_guard<std::mutex> lk(p->mu);
-  if (len)
+  if (len) {
     issue_copy(a->gpu_dst + offset, a->host + offset, len,
                hipMemcpyHostToDevice, p->h2d);
+    Telemetry::get().staged_h2d_bytes.fetch_add(len,
+                                                std::memory_order_relaxed);
+  }
   if (last) {
     a->done_ev = p->get_event();
     HIP_WARN(hipEventRecord(a->done_ev, p->h2d));

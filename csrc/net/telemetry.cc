@@ -1,0 +1,130 @@
+#include "telemetry.h"
+
+#include <time.h>
+
+#include <cstdio>
+#include <cstdlib>
+
+#include "baguanet/config.h"
+
+namespace baguanet {
+
+constexpr uint64_t Telemetry::kBounds[7];
+
+uint64_t now_ns() {
+  timespec ts;
+  clock_gettime(CLOCK_MONOTONIC, &ts);
+  return (uint64_t)ts.tv_sec * 1000000000ull + ts.tv_nsec;
+}
+
+Telemetry& Telemetry::get() {
+  static Telemetry* t = [] {
+    auto* p = new Telemetry();
+    p->spans_on = !Config::get().trace_file.empty();
+    atexit([] {
+      const Config& c = Config::get();
+      if (!c.metrics_file.empty())
+        Telemetry::get().dump_metrics(c.metrics_file.c_str());
+      if (!c.trace_file.empty())
+        Telemetry::get().dump_trace(c.trace_file.c_str());
+    });
+    return p;
+  }();
+  return *t;
+}
+
+void Telemetry::hist_add(std::atomic<uint64_t>* h, uint64_t bytes) {
+  int i = 0;
+  while (i < 7 && bytes > kBounds[i]) i++;
+  h[i].fetch_add(1, std::memory_order_relaxed);
+}
+
+uint32_t Telemetry::span_begin(uint8_t kind, uint64_t comm, uint32_t seq,
+                               uint32_t nbytes) {
+  if (!spans_on) return UINT32_MAX;
+  uint32_t slot = span_next.fetch_add(1, std::memory_order_relaxed) %
+                  kSpanCap;
+  Span& s = spans[slot];
+  s.t0 = now_ns();
+  s.t1 = 0;
+  s.comm = comm;
+  s.seq = seq;
+  s.nbytes = nbytes;
+  s.kind = kind;
+  return slot;
+}
+
+void Telemetry::span_end(uint32_t slot) {
+  if (slot == UINT32_MAX) return;
+  spans[slot % kSpanCap].t1 = now_ns();
+}
+
+void Telemetry::dump_metrics(const char* path) {
+  FILE* f = fopen(path, "w");
+  if (!f) return;
+  fprintf(f, "# baguanet transport metrics (prometheus text format)\n");
+  fprintf(f, "bnet_isend_total %llu\n",
+          (unsigned long long)isend_count.load());
+  fprintf(f, "bnet_irecv_total %llu\n",
+          (unsigned long long)irecv_count.load());
+  fprintf(f, "bnet_bytes_sent_total %llu\n",
+          (unsigned long long)bytes_sent.load());
+  fprintf(f, "bnet_bytes_recv_total %llu\n",
+          (unsigned long long)bytes_recv.load());
+  fprintf(f, "bnet_send_comms_total %llu\n",
+          (unsigned long long)send_comms.load());
+  fprintf(f, "bnet_recv_comms_total %llu\n",
+          (unsigned long long)recv_comms.load());
+  fprintf(f, "bnet_staged_d2h_bytes_total %llu\n",
+          (unsigned long long)staged_d2h_bytes.load());
+  fprintf(f, "bnet_staged_h2d_bytes_total %llu\n",
+          (unsigned long long)staged_h2d_bytes.load());
+  const char* names[2] = {"bnet_isend_nbytes", "bnet_irecv_nbytes"};
+  std::atomic<uint64_t>* hists[2] = {isend_hist, irecv_hist};
+  for (int h = 0; h < 2; h++) {
+    uint64_t cum = 0;
+    for (int i = 0; i < 7; i++) {
+      cum += hists[h][i].load();
+      fprintf(f, "%s_bucket{le=\"%llu\"} %llu\n", names[h],
+              (unsigned long long)kBounds[i], (unsigned long long)cum);
+    }
+    cum += hists[h][7].load();
+    fprintf(f, "%s_bucket{le=\"+Inf\"} %llu\n", names[h],
+            (unsigned long long)cum);
+  }
+  fclose(f);
+}
+
+void Telemetry::dump_trace(const char* path) {
+  FILE* f = fopen(path, "w");
+  if (!f) return;
+  fprintf(f, "[\n");
+  uint32_t n = span_next.load();
+  uint32_t count = n < kSpanCap ? n : kSpanCap;
+  bool first = true;
+  for (uint32_t i = 0; i < count; i++) {
+    const Span& s = spans[i];
+    if (s.t1 == 0 || s.t1 < s.t0) continue;
+    if (!first) fprintf(f, ",\n");
+    first = false;
+    fprintf(f,
+            "{\"name\":\"%s seq=%u %uB\",\"ph\":\"X\",\"pid\":1,"
+            "\"tid\":%llu,\"ts\":%.3f,\"dur\":%.3f}",
+            s.kind == 0 ? "isend" : "irecv", s.seq, s.nbytes,
+            (unsigned long long)(s.comm & 0xffff), s.t0 / 1000.0,
+            (s.t1 - s.t0) / 1000.0);
+  }
+  fprintf(f, "\n]\n");
+  fclose(f);
+}
+
+}  // namespace baguanet
+
+extern "C" {
+__attribute__((visibility("default"))) void bnet_dump_metrics(const char* path) {
+  baguanet::Telemetry::get().dump_metrics(path);
+}
+__attribute__((visibility("default"))) void bnet_dump_trace(const char* path) {
+  baguanet::Telemetry::get().dump_trace(path);
+}
+}

@@ -445,9 +445,18 @@ void Engine::unregister_sock_sync(TcpSock* s) {
   if (s->io_thread >= 0) threads_[s->io_thread].remove_sock_sync(s);
 }
 
-void Engine::kick_comm(SendComm* c) {
-  for (TcpSock* s : c->socks)
-    if (!s->tx.active && !s->want_epollout) threads_[s->io_thread].kick(s);
+void Engine::kick_comm(SendComm* c, int max_socks) {
+  // Kicking more sockets than the message has chunks just burns eventfd
+  // wakeups (hurts small-message rates); idle sockets beyond the first
+  // chunk-count workers contribute nothing.
+  int n = max_socks < 0 ? (int)c->socks.size() : max_socks;
+  for (TcpSock* s : c->socks) {
+    if (n <= 0) break;
+    if (!s->tx.active && !s->want_epollout) {
+      threads_[s->io_thread].kick(s);
+      n--;
+    }
+  }
 }
 
 void Engine::kick_comm(RecvComm* c) {

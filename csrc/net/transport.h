@@ -104,6 +104,7 @@ struct SendRequest {
   std::atomic<bool> hdr_sent{false};
   StageAlloc* stage = nullptr;  // non-null for NCCL_PTR_CUDA sends
   SendComm* comm = nullptr;
+  uint32_t span_slot = UINT32_MAX;
 
   bool complete() const {
     return total == 0 ? hdr_sent.load(std::memory_order_acquire)
@@ -121,6 +122,7 @@ struct RecvRequest {
   std::atomic<bool> gpu_done{false};   // H2D staging drained (CUDA dst)
   StageAlloc* stage = nullptr;         // bounce buffer for CUDA recv
   RecvComm* comm = nullptr;
+  uint32_t span_slot = UINT32_MAX;
 
   bool socket_complete() const {
     int64_t t = total.load(std::memory_order_acquire);
@@ -267,7 +269,7 @@ class Engine {
   int nthreads() const { return (int)threads_.size(); }
   void register_sock(TcpSock* s);
   void unregister_sock_sync(TcpSock* s);
-  void kick_comm(SendComm* c);
+  void kick_comm(SendComm* c, int max_socks = -1);
   void kick_comm(RecvComm* c);
 
  private:

@@ -1,0 +1,69 @@
+"""Telemetry: counters and spans must reflect actual transfers (run in a
+subprocess so env + atexit dumps are isolated)."""
+
+import json
+import multiprocessing as mp
+import os
+
+
+def _run(tmpdir, q):
+    os.environ["NCCL_SOCKET_IFNAME"] = "lo"
+    os.environ["BNET_TRACE_FILE"] = os.path.join(tmpdir, "trace.json")
+    os.environ["BNET_METRICS_FILE"] = os.path.join(tmpdir, "metrics.prom")
+    import ctypes as C
+
+    from baguanet.plugin import Plugin
+
+    import sys
+    sys.path.insert(0, os.path.join(os.path.dirname(__file__)))
+    from test_plugin_loopback import establish, xfer
+
+    p = Plugin()
+    lcomm, scomm, rcomm = establish(p)
+    for size in (100, 5000, 200000):
+        payload = os.urandom(size)
+        assert xfer(p, scomm, rcomm, payload) == payload
+    # on-demand dump through the exported C API
+    mfile = os.path.join(tmpdir, "ondemand.prom")
+    p.lib.bnet_dump_metrics(mfile.encode())
+    p.lib.bnet_dump_trace(
+        os.path.join(tmpdir, "ondemand_trace.json").encode()
+    )
+    p.close_send(scomm)
+    p.close_recv(rcomm)
+    p.close_listen(lcomm)
+    q.put("ok")
+
+
+def test_metrics_and_spans(tmp_path):
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    proc = ctx.Process(target=_run, args=(str(tmp_path), q))
+    proc.start()
+    assert q.get(timeout=120) == "ok"
+    proc.join(30)
+    assert proc.exitcode == 0
+
+    metrics = (tmp_path / "ondemand.prom").read_text()
+    m = {}
+    for line in metrics.splitlines():
+        if line.startswith("#") or not line.strip():
+            continue
+        k, v = line.rsplit(" ", 1)
+        m[k] = int(v)
+    assert m["bnet_isend_total"] == 3
+    assert m["bnet_irecv_total"] == 3
+    assert m["bnet_bytes_sent_total"] == 100 + 5000 + 200000
+    assert m["bnet_bytes_recv_total"] == 100 + 5000 + 200000
+    assert m["bnet_send_comms_total"] == 1
+    assert m['bnet_isend_nbytes_bucket{le="+Inf"}'] == 3
+
+    spans = json.loads((tmp_path / "ondemand_trace.json").read_text())
+    isends = [s for s in spans if s["name"].startswith("isend")]
+    irecvs = [s for s in spans if s["name"].startswith("irecv")]
+    assert len(isends) == 3 and len(irecvs) == 3
+    assert all(s["dur"] > 0 for s in spans)
+
+    # atexit dumps also fired
+    assert (tmp_path / "metrics.prom").exists()
+    assert (tmp_path / "trace.json").exists()
