@@ -121,6 +121,20 @@ void Telemetry::dump_trace(const char* path) {
 }  // namespace baguanet
 
 extern "C" {
+// Live config snapshot as JSON (tests verify env knobs are honored).
+__attribute__((visibility("default"))) int bnet_config_json(char* buf,
+                                                            int len) {
+  const baguanet::Config& c = baguanet::Config::get();
+  return snprintf(buf, (size_t)len,
+                  "{\"nstreams\":%d,\"min_chunk\":%u,\"max_chunk\":%u,"
+                  "\"io_threads\":%d,\"sockbuf\":%d,\"cuda_ptr\":%d,"
+                  "\"stage_pool\":%zu,\"stage_chunk\":%u,"
+                  "\"stage_kernel\":%d,\"backlog\":%d}",
+                  c.nstreams, c.min_chunk, c.max_chunk, c.io_threads,
+                  c.sockbuf, (int)c.cuda_ptr, c.stage_pool, c.stage_chunk,
+                  c.stage_kernel, c.backlog);
+}
+
 __attribute__((visibility("default"))) void bnet_dump_metrics(const char* path) {
   baguanet::Telemetry::get().dump_metrics(path);
 }

@@ -56,7 +56,8 @@ struct Telemetry {
 }  // namespace baguanet
 
 extern "C" {
-// on-demand dumps (used by tests via ctypes)
+// on-demand dumps + config introspection (used by tests via ctypes)
 void bnet_dump_metrics(const char* path);
 void bnet_dump_trace(const char* path);
+int bnet_config_json(char* buf, int len);
 }

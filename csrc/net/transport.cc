@@ -335,8 +335,9 @@ void IoThread::progress_recv(TcpSock* s) {
             return;
           }
         }
-        c->error.store(errno ? errno : ECONNRESET);
-        BNET_WARN("bnet recv socket error/eof: %s", strerror(errno));
+        c->error.store(n == 0 ? ECONNRESET : (errno ? errno : EIO));
+        BNET_WARN("bnet recv socket %s", n == 0 ? "eof mid-protocol"
+                                                : strerror(errno));
         return;
       }
       // header complete → locate the posted request
@@ -393,8 +394,9 @@ void IoThread::progress_recv(TcpSock* s) {
       }
       if (n < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) return;
       if (n < 0 && errno == EINTR) continue;
-      c->error.store(errno ? errno : ECONNRESET);
-      BNET_WARN("bnet recv payload error/eof: %s", strerror(errno));
+      c->error.store(n == 0 ? ECONNRESET : (errno ? errno : EIO));
+      BNET_WARN("bnet recv payload %s", n == 0 ? "eof mid-chunk"
+                                               : strerror(errno));
       return;
     }
     finish_chunk(s);

@@ -1,0 +1,147 @@
+"""Config env knobs and failure-path behavior (subprocess-isolated)."""
+
+import json
+import multiprocessing as mp
+import os
+
+
+def _config_probe(env, q):
+    for k, v in env.items():
+        os.environ[k] = v
+    import ctypes as C
+
+    from baguanet.plugin import Plugin
+
+    p = Plugin()
+    buf = C.create_string_buffer(1024)
+    p.lib.bnet_config_json(buf, 1024)
+    q.put(buf.value.decode())
+
+
+def _run_sub(target, env):
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    proc = ctx.Process(target=target, args=(env, q))
+    proc.start()
+    out = q.get(timeout=120)
+    proc.join(30)
+    assert proc.exitcode == 0
+    return out
+
+
+def test_env_knobs_respected():
+    cfg = json.loads(
+        _run_sub(
+            _config_probe,
+            {
+                "NCCL_SOCKET_IFNAME": "lo",
+                "BNET_NSTREAMS": "7",
+                "BNET_MIN_CHUNKSIZE": "65536",
+                "BNET_MAX_CHUNKSIZE": "2097152",
+                "BNET_IO_THREADS": "3",
+                "BNET_STAGE_KERNEL": "1",
+            },
+        )
+    )
+    assert cfg["nstreams"] == 7
+    assert cfg["min_chunk"] == 65536
+    assert cfg["max_chunk"] == 2097152
+    assert cfg["io_threads"] == 3
+    assert cfg["stage_kernel"] == 1
+
+
+def test_config_clamps():
+    cfg = json.loads(
+        _run_sub(
+            _config_probe,
+            {
+                "NCCL_SOCKET_IFNAME": "lo",
+                "BNET_NSTREAMS": "0",       # clamped to 1
+                "BNET_MIN_CHUNKSIZE": "17",  # clamped to 4096
+                "BNET_MAX_CHUNKSIZE": "1",   # clamped to >= min
+            },
+        )
+    )
+    assert cfg["nstreams"] == 1
+    assert cfg["min_chunk"] == 4096
+    assert cfg["max_chunk"] >= cfg["min_chunk"]
+
+
+def _peer_death_receiver(env, q):
+    for k, v in env.items():
+        os.environ[k] = v
+    import ctypes as C
+    import sys
+    import time
+
+    sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+    from baguanet.plugin import Plugin
+
+    p = Plugin()
+    handle, lcomm = p.listen(0)
+    q.put(bytes(handle))
+    rcomm = None
+    t0 = time.monotonic()
+    while rcomm is None and time.monotonic() - t0 < 60:
+        rcomm = p.accept(lcomm)
+    assert rcomm is not None
+    # wait until the sender is already dead before posting, so only the
+    # few MB the kernel buffered exist and EOF arrives mid-message
+    time.sleep(2.0)
+    buf = C.create_string_buffer(256 << 20)
+    mh = p.reg_mr(rcomm, buf, 256 << 20)
+    req = p.irecv(rcomm, buf, 256 << 20, mh)
+    q.put("posted")
+    # the peer dies mid-protocol; test() must surface an error, not hang
+    t0 = time.monotonic()
+    while time.monotonic() - t0 < 60:
+        try:
+            done, _ = p.test(req)
+        except RuntimeError as e:
+            q.put(f"error-surfaced:{e}")
+            return
+        if done:
+            q.put("unexpected-completion")
+            return
+        time.sleep(0.01)
+    q.put("timeout-no-error")
+
+
+def _peer_death_sender(env, handle_bytes, q):
+    for k, v in env.items():
+        os.environ[k] = v
+    import ctypes as C
+
+    from baguanet.plugin import Plugin
+
+    p = Plugin()
+    handle = (C.c_char * len(handle_bytes)).from_buffer_copy(handle_bytes)
+    scomm = None
+    while scomm is None:
+        scomm = p.connect(0, handle)
+    # start a send, then die abruptly without completing the protocol
+    buf = C.create_string_buffer(256 << 20)
+    mh = p.reg_mr(scomm, buf, 256 << 20)
+    p.isend(scomm, buf, 256 << 20, mh)
+    q.put("sender-started")
+    import time
+
+    time.sleep(0.3)  # let the queue feeder thread flush before hard exit
+    os._exit(42)  # hard kill: sockets reset mid-message
+
+
+def test_peer_death_surfaces_error():
+    env = {"NCCL_SOCKET_IFNAME": "lo", "BNET_NSTREAMS": "2"}
+    ctx = mp.get_context("spawn")
+    qr, qs = ctx.Queue(), ctx.Queue()
+    pr = ctx.Process(target=_peer_death_receiver, args=(env, qr))
+    pr.start()
+    handle = qr.get(timeout=60)
+    ps = ctx.Process(target=_peer_death_sender, args=(env, handle, qs))
+    ps.start()
+    assert qr.get(timeout=60) == "posted"
+    assert qs.get(timeout=60) == "sender-started"
+    ps.join(30)
+    result = qr.get(timeout=90)
+    pr.join(30)
+    assert result.startswith("error-surfaced"), result
