@@ -85,7 +85,11 @@ class BucketedDDP(nn.Module):
         cur_bytes = 0
         for p in reversed(params):
             sz = p.numel() * p.element_size()
-            if cur and cur_bytes + sz > cap:
+            # a bucket is one flat tensor: same dtype + device throughout
+            mismatch = cur and (
+                p.dtype != cur[-1].dtype or p.device != cur[-1].device
+            )
+            if cur and (cur_bytes + sz > cap or mismatch):
                 self.buckets.append(self._make_bucket(cur))
                 cur, cur_bytes = [], 0
             cur.append(p)

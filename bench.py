@@ -75,21 +75,24 @@ def main():
     from baguanet.parallel import BucketedDDP
 
     torch.manual_seed(42 + rank)
-    dtype = torch.float32 if args.dtype == "fp32" else torch.bfloat16
+    use_bf16 = args.dtype == "bf16"
     model = (vgg16() if args.model == "vgg16" else resnet50()).to(device)
-    if dtype != torch.float32:
-        model = model.to(dtype)
     model = BucketedDDP(model, bucket_cap_mb=50.0)
     opt = torch.optim.SGD(model.module.parameters(), lr=0.01, momentum=0.9)
 
-    x = torch.randn(args.batch, 3, 224, 224, device=device, dtype=dtype)
+    x = torch.randn(args.batch, 3, 224, 224, device=device)
     y = torch.randint(0, 1000, (args.batch,), device=device)
     loss_fn = torch.nn.CrossEntropyLoss()
 
     def step():
         model.zero_grad()
-        out = model(x)
-        loss = loss_fn(out.float(), y)
+        # bf16 = autocast compute with fp32 params/grads (BASELINE config 5
+        # "ResNet-50 bf16 DDP-style allreduce"); fp32 = the reference
+        # synthetic_benchmark's dtype
+        with torch.autocast(device_type="cuda", dtype=torch.bfloat16,
+                            enabled=use_bf16 and use_cuda):
+            out = model(x)
+            loss = loss_fn(out.float(), y)
         loss.backward()
         model.finish_backward()
         opt.step()

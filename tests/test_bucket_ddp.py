@@ -129,3 +129,22 @@ def test_no_sync_and_world1():
         loss3.backward()
     for p, rp in zip(model.module.parameters(), ref.parameters()):
         torch.testing.assert_close(p.grad, 2 * rp.grad, rtol=1e-5, atol=1e-6)
+
+
+def test_mixed_dtype_buckets():
+    """Buckets must split on dtype boundaries (flat buffer is one dtype)."""
+    import torch
+    from baguanet.parallel import BucketedDDP
+
+    class M(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.a = torch.nn.Linear(8, 8)
+            self.b = torch.nn.Linear(8, 8).to(torch.bfloat16)
+            self.c = torch.nn.Linear(8, 8)
+
+    m = BucketedDDP(M(), bucket_cap_mb=100, broadcast_params=False)
+    for bkt in m.buckets:
+        dts = {p.dtype for p in bkt.params}
+        assert len(dts) == 1
+        assert bkt.flat.dtype == next(iter(dts))
