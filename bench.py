@@ -63,9 +63,6 @@ def main():
     device = torch.device(f"cuda:{local_rank}" if use_cuda else "cpu")
     if use_cuda:
         torch.cuda.set_device(device)
-        # MIOpen kernel search for the fixed conv shapes (synthetic bench
-        # reuses one shape set; search cost is paid in warmup)
-        torch.backends.cudnn.benchmark = True
     if world > 1:
         dist.init_process_group(
             "nccl" if use_cuda else "gloo", rank=rank, world_size=world

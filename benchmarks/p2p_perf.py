@@ -60,10 +60,10 @@ def _receiver(conn, args, out_q):
                     break
                 reqs.append(r)
                 posted += 1
-            if reqs:
-                ok, _ = p.test(reqs[0])
+            for r in list(reqs):
+                ok, _ = p.test(r)
                 if ok:
-                    reqs.pop(0)
+                    reqs.remove(r)
                     done += 1
         conn.send(("size-done", size))
     conn.recv()
@@ -100,10 +100,10 @@ def _sender(conn, args, out_q):
                     break
                 reqs.append(r)
                 posted += 1
-            if reqs:
-                ok, _ = p.test(reqs[0])
+            for r in list(reqs):
+                ok, _ = p.test(r)
                 if ok:
-                    reqs.pop(0)
+                    reqs.remove(r)
                     done += 1
         # wait for receiver to fully drain this size
         tag, s = conn.recv()

@@ -15,7 +15,7 @@
 namespace baguanet {
 
 constexpr int kBlock = 256;
-constexpr uint32_t kTileBytes = 64 * 1024;  // one work item moves ≤64 KiB
+constexpr uint32_t kTileBytes = 256 * 1024;  // one work item moves ≤256 KiB
 constexpr int kMaxDesc = 512;               // descriptors per launch
 
 struct PackDesc {

@@ -39,7 +39,11 @@ struct StageAlloc {
   SendRequest* sreq = nullptr;
   // recv pipeline
   char* gpu_dst = nullptr;
-  hipEvent_t done_ev = nullptr;
+  StagePool* pool = nullptr;
+  // published (release) only AFTER hipEventRecord — readers load acquire;
+  // a cached/reused event queried before its re-record reports complete,
+  // so visibility must imply recorded
+  std::atomic<hipEvent_t> done_ev{nullptr};
 };
 
 class StagePool {
@@ -222,6 +226,7 @@ bool stage_recv_begin(StagePool* p, RecvRequest* req, void* dst,
   }
   a->size = std::max(capacity, 1u);
   a->gpu_dst = (char*)dst;
+  a->pool = p;
   req->stage = a;
   return true;
 }
@@ -230,28 +235,34 @@ char* stage_recv_base(RecvRequest* req) {
   return req->stage ? ((StageAlloc*)req->stage)->host : req->dst;
 }
 
-void stage_recv_chunk(RecvRequest* req, uint32_t offset, uint32_t len,
-                      bool last) {
+void stage_recv_issue(RecvRequest* req, uint32_t offset, uint32_t len) {
   StageAlloc* a = (StageAlloc*)req->stage;
-  StagePool* p = req->comm->stage_pool;
+  StagePool* p = a->pool;
+  if (!len) return;
   std::lock_guard<std::mutex> lk(p->mu);
-  if (len) {
-    issue_copy(a->gpu_dst + offset, a->host + offset, len,
-               hipMemcpyHostToDevice, p->h2d);
-    Telemetry::get().staged_h2d_bytes.fetch_add(len,
-                                                std::memory_order_relaxed);
+  issue_copy(a->gpu_dst + offset, a->host + offset, len,
+             hipMemcpyHostToDevice, p->h2d);
+  Telemetry::get().staged_h2d_bytes.fetch_add(len, std::memory_order_relaxed);
+}
+
+void stage_recv_last(RecvRequest* req) {
+  StageAlloc* a = (StageAlloc*)req->stage;
+  StagePool* p = a->pool;
+  hipEvent_t ev;
+  {
+    std::lock_guard<std::mutex> lk(p->mu);
+    ev = p->get_event();
+    HIP_WARN(hipEventRecord(ev, p->h2d));
   }
-  if (last) {
-    a->done_ev = p->get_event();
-    HIP_WARN(hipEventRecord(a->done_ev, p->h2d));
-  }
+  a->done_ev.store(ev, std::memory_order_release);
 }
 
 bool stage_recv_done(RecvRequest* req) {
   StageAlloc* a = (StageAlloc*)req->stage;
   if (!a) return true;
-  if (!a->done_ev) return false;  // last chunk not yet landed
-  return hipEventQuery(a->done_ev) == hipSuccess;
+  hipEvent_t ev = a->done_ev.load(std::memory_order_acquire);
+  if (!ev) return false;  // last chunk not yet issued+recorded
+  return hipEventQuery(ev) == hipSuccess;
 }
 
 void stage_release(StagePool* p, SendRequest* req) {
@@ -267,7 +278,8 @@ void stage_release(StagePool* p, RecvRequest* req) {
   StageAlloc* a = (StageAlloc*)req->stage;
   if (!a) return;
   std::lock_guard<std::mutex> lk(p->mu);
-  if (a->done_ev) p->put_event(a->done_ev);
+  hipEvent_t ev = a->done_ev.load(std::memory_order_relaxed);
+  if (ev) p->put_event(ev);
   p->free(a->pool_off, a->size);
   req->stage = nullptr;
   delete a;

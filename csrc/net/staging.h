@@ -62,9 +62,11 @@ bool stage_recv_begin(StagePool* p, RecvRequest* req, void* dst,
 char* stage_recv_base(RecvRequest* req);
 
 // A chunk [offset, offset+len) just finished on a socket: issue its H2D
-// copy; when `last` also record the completion event.
-void stage_recv_chunk(RecvRequest* req, uint32_t offset, uint32_t len,
-                      bool last);
+// copy.  Callers must issue BEFORE counting the chunk as received.
+void stage_recv_issue(RecvRequest* req, uint32_t offset, uint32_t len);
+
+// All chunks are issued (received == total): record the completion event.
+void stage_recv_last(RecvRequest* req);
 
 // True when all H2D copies for the request have drained (event landed).
 bool stage_recv_done(RecvRequest* req);

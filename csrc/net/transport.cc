@@ -412,9 +412,13 @@ void IoThread::finish_chunk(TcpSock* s) {
   s->rx.hdr_got = 0;
   s->rx.req = nullptr;
   c->stats.bytes_recv.fetch_add(h.len, std::memory_order_relaxed);
+  // ORDER MATTERS: issue this chunk's H2D copy BEFORE counting it, so when
+  // any thread observes received == total, every chunk's copy has been
+  // submitted and the completion event recorded after that covers them all.
+  if (r->stage) stage_recv_issue(r, h.offset, h.len);
   uint32_t got = r->received.fetch_add(h.len, std::memory_order_acq_rel) +
                  h.len;
-  if (r->stage) stage_recv_chunk(r, h.offset, h.len, got == h.total);
+  if (r->stage && got == h.total) stage_recv_last(r);
 }
 
 // -------------------------------------------------------------- Engine ----
