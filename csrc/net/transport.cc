@@ -17,6 +17,7 @@
 
 #include "baguanet/log.h"
 #include "staging.h"
+#include "telemetry.h"
 
 namespace baguanet {
 
@@ -148,18 +149,26 @@ void IoThread::run() {
   snprintf(tname, sizeof(tname), "bnet-io%d", idx_);
   pthread_setname_np(pthread_self(), tname);
   epoll_event evs[64];
+  uint64_t last_active_ns = 0;
   while (!stop_.load(std::memory_order_relaxed)) {
-    // Spin (timeout 0) while any socket has staging copies pending so GPU
-    // watermarks advance promptly; otherwise block.
+    // Spin (timeout 0) while staging copies are pending (GPU watermarks
+    // must advance promptly) or within a short window after traffic —
+    // a blocked epoll_wait costs an eventfd wakeup (~5-10 us) per
+    // message, which dominates small-message rates.  NCCL's own proxy
+    // threads spin the same way.
     bool staging_busy = false;
     for (TcpSock* s : socks_) {
       if (s->scomm && s->scomm->stage_pool &&
-          stage_pending(s->scomm->stage_pool))
+          stage_pending(s->scomm->stage_pool)) {
+        stage_poll(s->scomm->stage_pool);
         staging_busy = true;
+      }
     }
-    int timeout = staging_busy ? 0 : 100;
-    int n = epoll_wait(epfd_, evs, 64, timeout);
+    uint64_t now = now_ns();
+    bool spin = staging_busy || (now - last_active_ns < 200'000);
+    int n = epoll_wait(epfd_, evs, 64, spin ? 0 : 100);
     if (n < 0 && errno != EINTR) break;
+    if (n > 0) last_active_ns = now;
     bool had_ev = false;
     for (int i = 0; i < n; i++) {
       if (evs[i].data.ptr == nullptr) {
@@ -170,9 +179,9 @@ void IoThread::run() {
       }
       progress(static_cast<TcpSock*>(evs[i].data.ptr));
     }
-    if (had_ev || staging_busy) handle_tasks();
-    if (staging_busy) {
-      // watermarks may have advanced: retry idle senders
+    if (had_ev || spin) handle_tasks();
+    if (spin) {
+      // watermarks/jobs may have advanced without an epoll event
       for (TcpSock* s : socks_)
         if (!s->is_recv && !s->tx.active && !s->want_epollout) progress(s);
     }
