@@ -199,6 +199,9 @@ def rccl_env(
     benchmarking of the transport itself; normal runs keep xGMI intra-node).
     """
     e = dict(env if env is not None else os.environ)
+    # RCCL resolves the plugin as librccl-net-<name>.so (observed on the
+    # MI355X box: "NET/Plugin: Could not find: librccl-net-bagua.so");
+    # the build dir ships librccl-net-bagua.so + libnccl-net-bagua.so.
     e["NCCL_NET_PLUGIN"] = "bagua"
     e["LD_LIBRARY_PATH"] = (
         f"{PLUGIN_DIR}:{e.get('LD_LIBRARY_PATH', '')}".rstrip(":")

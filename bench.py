@@ -40,6 +40,8 @@ def main():
     ap.add_argument("--dtype", default="fp32", choices=["fp32", "bf16"])
     ap.add_argument("--force-net", action="store_true",
                     help="route even intra-node traffic through the plugin")
+    ap.add_argument("--channels-last", action="store_true",
+                    help="NHWC memory format (MIOpen igemm path)")
     args = ap.parse_args()
 
     # plugin env must be set before the first collective
@@ -74,10 +76,14 @@ def main():
     torch.manual_seed(42 + rank)
     use_bf16 = args.dtype == "bf16"
     model = (vgg16() if args.model == "vgg16" else resnet50()).to(device)
+    if args.channels_last:
+        model = model.to(memory_format=torch.channels_last)
     model = BucketedDDP(model, bucket_cap_mb=50.0)
     opt = torch.optim.SGD(model.module.parameters(), lr=0.01, momentum=0.9)
 
     x = torch.randn(args.batch, 3, 224, 224, device=device)
+    if args.channels_last:
+        x = x.to(memory_format=torch.channels_last)
     y = torch.randint(0, 1000, (args.batch,), device=device)
     loss_fn = torch.nn.CrossEntropyLoss()
 

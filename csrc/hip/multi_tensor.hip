@@ -15,7 +15,8 @@
 namespace baguanet {
 
 constexpr int kBlock = 256;
-constexpr uint32_t kTileBytes = 256 * 1024;  // one work item moves ≤256 KiB
+constexpr uint32_t kMaxTile = 256 * 1024;   // upper bound per work item
+constexpr uint32_t kMinTile = 64 * 1024;    // lower bound per work item
 constexpr int kMaxDesc = 512;               // descriptors per launch
 
 struct PackDesc {
@@ -58,12 +59,12 @@ __device__ inline void copy_span(char* __restrict__ dst,
 
 __global__ void multi_copy_kernel(const PackDesc* __restrict__ descs,
                                   const WorkItem* __restrict__ items,
-                                  uint32_t nitems) {
+                                  uint32_t nitems, uint32_t tile_bytes) {
   for (uint32_t w = blockIdx.x; w < nitems; w += gridDim.x) {
     WorkItem it = items[w];
     PackDesc d = descs[it.desc];
-    uint32_t off = (uint32_t)it.tile * kTileBytes;
-    uint32_t n = d.bytes - off < kTileBytes ? d.bytes - off : kTileBytes;
+    uint32_t off = (uint32_t)it.tile * tile_bytes;
+    uint32_t n = d.bytes - off < tile_bytes ? d.bytes - off : tile_bytes;
     copy_span((char*)d.dst + off, (const char*)d.src + off, n);
   }
 }
@@ -74,10 +75,17 @@ __global__ void multi_copy_kernel(const PackDesc* __restrict__ descs,
 size_t multi_copy_launch(const PackDesc* host_descs, int ndesc, void* scratch,
                          size_t scratch_bytes, void* staging,
                          hipStream_t stream) {
+  // Pick the tile so the launch has >= ~2048 work items (fills 256 CUs x
+  // several blocks) without exceeding the per-item upper bound.
+  size_t total = 0;
+  for (int i = 0; i < ndesc; i++) total += host_descs[i].bytes;
+  uint32_t tile = kMaxTile;
+  while (tile > kMinTile && total / tile < 2048) tile /= 2;
+
   // layout: [ndesc PackDesc][nitems WorkItem]
   uint32_t nitems = 0;
   for (int i = 0; i < ndesc; i++)
-    nitems += (host_descs[i].bytes + kTileBytes - 1) / kTileBytes;
+    nitems += (host_descs[i].bytes + tile - 1) / tile;
   size_t need = sizeof(PackDesc) * ndesc + sizeof(WorkItem) * nitems;
   if (need > scratch_bytes || ndesc > kMaxDesc) return need;
 
@@ -86,7 +94,7 @@ size_t multi_copy_launch(const PackDesc* host_descs, int ndesc, void* scratch,
   WorkItem* hitems = (WorkItem*)(h + sizeof(PackDesc) * ndesc);
   uint32_t w = 0;
   for (int i = 0; i < ndesc; i++) {
-    uint32_t t = (host_descs[i].bytes + kTileBytes - 1) / kTileBytes;
+    uint32_t t = (host_descs[i].bytes + tile - 1) / tile;
     for (uint32_t j = 0; j < t; j++) hitems[w++] = {(uint16_t)i, (uint16_t)j};
   }
   (void)hipMemcpyAsync(scratch, staging, need, hipMemcpyHostToDevice, stream);
@@ -95,7 +103,7 @@ size_t multi_copy_launch(const PackDesc* host_descs, int ndesc, void* scratch,
       (const WorkItem*)((char*)scratch + sizeof(PackDesc) * ndesc);
   uint32_t grid = nitems < 4096 ? (nitems ? nitems : 1) : 4096;
   hipLaunchKernelGGL(multi_copy_kernel, dim3(grid), dim3(kBlock), 0, stream,
-                     ddescs, ditems, nitems);
+                     ddescs, ditems, nitems, tile);
   return 0;
 }
 

@@ -59,12 +59,8 @@ def test_rccl_with_plugin_env(tmp_path):
     assert "RCCL_OK" in out
     # our v6 symbol must never be reported missing
     assert "Failed to find ncclNetPlugin_v6" not in out
-    plugin_lines = [
-        ln for ln in out.splitlines()
-        if "nccl-net" in ln or "NET/Plugin" in ln or "BaguaNetAMD" in ln
-    ]
-    for ln in plugin_lines:
-        low = ln.lower()
-        assert "error" not in low and "unable" not in low, (
-            "plugin load problem: " + ln
-        )
+    # RCCL resolves librccl-net-bagua.so and must have loaded v6 (net init
+    # runs even at world=1: topology probing enumerates net devices)
+    loaded = [ln for ln in out.splitlines() if "Loaded net plugin" in ln]
+    assert loaded, "RCCL did not load the plugin:\n" + out[-3000:]
+    assert any("v6" in ln for ln in loaded), loaded
