@@ -119,6 +119,71 @@ def _sender(conn, args, out_q):
     out_q.put(results)
 
 
+def _duplex_worker(conn, args, out_q, is_a):
+    """Each process sends AND receives simultaneously (ring-edge pattern)."""
+    _set_env(args)
+    from baguanet.plugin import Plugin
+
+    p = Plugin()
+    handle, lcomm = p.listen(0)
+    conn.send(bytes(handle))
+    peer_hb = conn.recv()
+    peer_handle = (C.c_char * len(peer_hb)).from_buffer_copy(peer_hb)
+    scomm = rcomm = None
+    while scomm is None or rcomm is None:
+        if scomm is None:
+            scomm = p.connect(0, peer_handle)
+        if rcomm is None:
+            rcomm = p.accept(lcomm)
+    smh = p.reg_mr(scomm, None, 0)
+    rmh = p.reg_mr(rcomm, None, 0)
+    results = []
+    for size in args.sizes:
+        n_msgs = max(4, min(args.max_msgs, args.bytes_per_size // size))
+        sbuf = C.create_string_buffer(os.urandom(size), size)
+        rbufs = [C.create_string_buffer(size) for _ in range(DEPTH)]
+        conn.send(("ready", size))
+        assert conn.recv() == ("ready", size)
+        t0 = time.perf_counter()
+        sdone = rdone = sposted = rposted = 0
+        sreqs, rreqs = [], []
+        while sdone < n_msgs or rdone < n_msgs:
+            while rposted < n_msgs and len(rreqs) < DEPTH:
+                r = p.irecv(rcomm, rbufs[rposted % DEPTH], size, rmh)
+                if r is None:
+                    break
+                rreqs.append(r)
+                rposted += 1
+            while sposted < n_msgs and len(sreqs) < DEPTH:
+                r = p.isend(scomm, sbuf, size, smh)
+                if r is None:
+                    break
+                sreqs.append(r)
+                sposted += 1
+            for r in list(sreqs):
+                if p.test(r)[0]:
+                    sreqs.remove(r)
+                    sdone += 1
+            for r in list(rreqs):
+                if p.test(r)[0]:
+                    rreqs.remove(r)
+                    rdone += 1
+        conn.send(("size-done", size))
+        assert conn.recv() == ("size-done", size)
+        dt = time.perf_counter() - t0
+        results.append(
+            {"size": size, "msgs": n_msgs, "secs": round(dt, 4),
+             "GBps_each_way": round(n_msgs * size / dt / 1e9, 3),
+             "GBps_aggregate": round(2 * n_msgs * size / dt / 1e9, 3)}
+        )
+    conn.send("done")
+    conn.recv()
+    p.close_send(scomm)
+    p.close_recv(rcomm)
+    p.close_listen(lcomm)
+    out_q.put(results if is_a else "b-ok")
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--nstreams", type=int, default=4)
@@ -130,13 +195,19 @@ def main():
     ap.add_argument("--bytes-per-size", type=int, default=2 << 30)
     ap.add_argument("--max-msgs", type=int, default=2000)
     ap.add_argument("--json", action="store_true")
+    ap.add_argument("--duplex", action="store_true",
+                    help="both directions simultaneously (ring-edge pattern)")
     args = ap.parse_args()
 
     ctx = mp.get_context("spawn")
     a, b = ctx.Pipe()
     q = ctx.Queue()
-    pr = ctx.Process(target=_receiver, args=(a, args, q))
-    ps = ctx.Process(target=_sender, args=(b, args, q))
+    if args.duplex:
+        pr = ctx.Process(target=_duplex_worker, args=(a, args, q, True))
+        ps = ctx.Process(target=_duplex_worker, args=(b, args, q, False))
+    else:
+        pr = ctx.Process(target=_receiver, args=(a, args, q))
+        ps = ctx.Process(target=_sender, args=(b, args, q))
     pr.start()
     ps.start()
     outs = [q.get(timeout=600), q.get(timeout=600)]
@@ -144,7 +215,7 @@ def main():
     ps.join(30)
     results = next(o for o in outs if isinstance(o, list))
     header = {
-        "bench": "plugin p2p one-way",
+        "bench": "plugin p2p duplex" if args.duplex else "plugin p2p one-way",
         "nstreams": args.nstreams,
         "io_threads": args.io_threads,
         "ifname": args.ifname,
@@ -154,8 +225,11 @@ def main():
     else:
         print(header)
         for r in results:
+            bw = r.get("GBps", r.get("GBps_aggregate"))
             print(f"  {r['size']:>10} B x {r['msgs']:>5} msgs: "
-                  f"{r['GBps']:8.3f} GB/s")
+                  f"{bw:8.3f} GB/s"
+                  + (" (aggregate both ways)" if "GBps_aggregate" in r
+                     else ""))
 
 
 if __name__ == "__main__":
