@@ -235,6 +235,7 @@ ncclResult_t Net::connect(int dev, void* handle, void** send_comm) {
   for (auto& p : t->socks) {
     auto* s = new TcpSock();
     s->fd = p.fd;
+    s->idx = (int)c->socks.size();
     s->is_recv = false;
     s->scomm = c;
     c->socks.push_back(s);
@@ -307,6 +308,7 @@ ncclResult_t Net::accept(void* listen_comm, void** recv_comm) {
     for (int fd : it->fds) {
       auto* s = new TcpSock();
       s->fd = fd;
+      s->idx = (int)c->socks.size();
       s->is_recv = true;
       s->rcomm = c;
       c->socks.push_back(s);
@@ -367,7 +369,11 @@ ncclResult_t Net::isend(void* send_comm, void* data, int size, int tag,
   r->span_slot = T.span_begin(0, (uint64_t)(uintptr_t)c, r->seq.load(),
                               (uint32_t)size);
   int nchunks = r->total ? (int)((r->total + r->chunk - 1) / r->chunk) : 1;
-  Engine::get().kick_comm(c, nchunks);
+  if (nchunks <= 1)
+    Engine::get().kick_sock(
+        c->socks[r->seq.load(std::memory_order_relaxed) % c->socks.size()]);
+  else
+    Engine::get().kick_comm(c, nchunks);
   *request = tag_send(r);
   return ncclSuccess;
 }

