@@ -40,8 +40,10 @@ def main():
     ap.add_argument("--dtype", default="fp32", choices=["fp32", "bf16"])
     ap.add_argument("--force-net", action="store_true",
                     help="route even intra-node traffic through the plugin")
-    ap.add_argument("--channels-last", action="store_true",
-                    help="NHWC memory format (MIOpen igemm path)")
+    ap.add_argument("--channels-last", action=argparse.BooleanOptionalAction,
+                    default=True,
+                    help="NHWC memory format (MIOpen igemm path; +9%% on "
+                         "VGG16 fp32, same numerics)")
     args = ap.parse_args()
 
     # plugin env must be set before the first collective
