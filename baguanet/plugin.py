@@ -187,6 +187,23 @@ class Plugin:
         self._check(self.vt.closeListen(lcomm), "closeListen")
 
 
+_preloaded = None
+
+
+def preload() -> None:
+    """dlopen the plugin with its SONAME registered, so RCCL's later
+    dlopen("librccl-net-bagua.so") resolves to the already-loaded library.
+
+    Needed because mutating LD_LIBRARY_PATH inside a running process does
+    not affect dlopen search paths (glibc snapshots it at startup) — the
+    torchrun-launched bench sets env in-process.
+    """
+    global _preloaded
+    if _preloaded is None:
+        _preloaded = C.CDLL(str(PLUGIN_DIR / "librccl-net-bagua.so"),
+                            mode=C.RTLD_GLOBAL)
+
+
 def rccl_env(
     env: Optional[dict] = None,
     ifname: Optional[str] = None,

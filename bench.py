@@ -46,9 +46,14 @@ def main():
                          "VGG16 fp32, same numerics)")
     args = ap.parse_args()
 
-    # plugin env must be set before the first collective
-    from baguanet.plugin import rccl_env
+    # plugin env must be set before the first collective; preload the .so
+    # so RCCL's dlopen resolves it regardless of startup LD_LIBRARY_PATH
+    from baguanet.plugin import preload, rccl_env
 
+    try:
+        preload()
+    except OSError:
+        pass  # plugin not built — bench still runs on stock RCCL
     for k, v in rccl_env(env={}, force_net=args.force_net).items():
         if k == "LD_LIBRARY_PATH":
             os.environ[k] = f"{v}:{os.environ.get(k, '')}".rstrip(":")

@@ -41,7 +41,12 @@ def main():
     args = ap.parse_args()
 
     if not args.no_plugin:
-        from baguanet.plugin import rccl_env
+        from baguanet.plugin import preload, rccl_env
+
+        try:
+            preload()
+        except OSError:
+            pass
 
         for k, v in rccl_env(env={}, force_net=args.force_net).items():
             if k == "LD_LIBRARY_PATH":
