@@ -162,7 +162,12 @@ ncclResult_t Net::connect(int dev, void* handle, void** send_comm) {
     t->socks.resize(cfg.nstreams);
     for (int i = 0; i < cfg.nstreams; i++) {
       int fd = socket(t->peer.ss_family, SOCK_STREAM | SOCK_CLOEXEC, 0);
-      if (fd < 0) return ncclSystemError;
+      if (fd < 0) {
+        for (auto& q : t->socks)
+          if (q.fd >= 0) close(q.fd);
+        delete t;
+        return ncclSystemError;
+      }
       tune_socket2(fd);
       // route via the chosen NIC
       if (dev >= 0 && dev < (int)devs_.size()) {
@@ -346,6 +351,11 @@ ncclResult_t Net::isend(void* send_comm, void* data, int size, int tag,
   r->hdr_sent.store(false, std::memory_order_relaxed);
   r->comm = c;
   if (ptr_type == NCCL_PTR_CUDA && size > 0) {
+    if ((size_t)size > Config::get().stage_pool) {
+      BNET_WARN("isend: %d B exceeds staging pool (%zu B) — raise "
+                "BNET_STAGE_POOL", size, Config::get().stage_pool);
+      return ncclInternalError;  // would retry forever otherwise
+    }
     if (!c->stage_pool) {
       c->stage_pool = stage_pool_create();
       if (!c->stage_pool) return ncclInternalError;
@@ -400,6 +410,11 @@ ncclResult_t Net::irecv(void* recv_comm, int n, void** data, int* sizes,
   r->gpu_done.store(false, std::memory_order_relaxed);
   r->comm = c;
   if (ptr_type == NCCL_PTR_CUDA) {
+    if ((size_t)r->capacity > Config::get().stage_pool) {
+      BNET_WARN("irecv: %u B exceeds staging pool (%zu B) — raise "
+                "BNET_STAGE_POOL", r->capacity, Config::get().stage_pool);
+      return ncclInternalError;
+    }
     if (!c->stage_pool) {
       c->stage_pool = stage_pool_create();
       if (!c->stage_pool) return ncclInternalError;
