@@ -379,9 +379,11 @@ ncclResult_t Net::isend(void* send_comm, void* data, int size, int tag,
   r->span_slot = T.span_begin(0, (uint64_t)(uintptr_t)c, r->seq.load(),
                               (uint32_t)size);
   int nchunks = r->total ? (int)((r->total + r->chunk - 1) / r->chunk) : 1;
-  if (nchunks <= 1)
+  if (nchunks <= 1) {
+    size_t nsmall = c->socks.size() < 2 ? c->socks.size() : 2;
     Engine::get().kick_sock(
-        c->socks[r->seq.load(std::memory_order_relaxed) % c->socks.size()]);
+        c->socks[r->seq.load(std::memory_order_relaxed) % nsmall]);
+  }
   else
     Engine::get().kick_comm(c, nchunks);
   *request = tag_send(r);

@@ -215,6 +215,11 @@ void IoThread::progress(TcpSock* s) {
 static SendRequest* claim_chunk(SendComm* c, int sock_idx, uint32_t* off,
                                 uint32_t* len) {
   int nsocks = (int)c->socks.size();
+  // Single-chunk messages fan out over at most 2 sockets: measured on the
+  // MI355X box, 64 KiB message rate is ~11 GB/s at 1-2 sockets but drops
+  // to 3-5 GB/s when spread over 4 (cross-thread handoff per message
+  // outweighs parallelism below the striping threshold).
+  int nsmall = nsocks < 2 ? nsocks : 2;
   uint32_t oldest = c->oldest.load(std::memory_order_acquire);
   uint32_t newest = oldest + NCCL_NET_MAX_REQUESTS;
   for (uint32_t s = oldest; s != newest; s++) {
@@ -235,7 +240,7 @@ static SendRequest* claim_chunk(SendComm* c, int sock_idx, uint32_t* off,
     }
     if (r->total <= r->chunk) {
       // single-chunk (or empty) message: statically owned
-      if ((int)(s % (uint32_t)nsocks) != sock_idx) continue;
+      if ((int)(s % (uint32_t)nsmall) != sock_idx) continue;
     }
     if (r->total == 0) {
       bool expect = false;
