@@ -271,6 +271,11 @@ static SendRequest* claim_chunk(SendComm* c, int sock_idx, uint32_t* off,
 void IoThread::progress_send(TcpSock* s) {
   SendComm* c = s->scomm;
   if (c->error.load(std::memory_order_relaxed)) return;
+  // Fairness credit: one dispatch writes at most `inflight_per_stream`
+  // bytes before yielding the IO thread to its other sockets (other
+  // comms sharing the NIC), then reschedules itself.
+  uint32_t budget = Config::get().inflight_per_stream;
+  if (budget < 65536) budget = 65536;
   while (true) {
     if (!s->tx.active) {
       uint32_t off = 0, len = 0;
@@ -327,6 +332,11 @@ void IoThread::progress_send(TcpSock* s) {
       c->stats.bytes_sent.fetch_add(len, std::memory_order_relaxed);
       r->sent.fetch_add(len, std::memory_order_acq_rel);
     }
+    if (len >= budget) {
+      kick(s);  // yield: requeue ourselves behind other sockets' work
+      return;
+    }
+    budget -= len;
   }
 }
 
