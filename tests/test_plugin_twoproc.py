@@ -22,7 +22,8 @@ def _payload(i, size):
 
 
 def _set_env():
-    os.environ["NCCL_SOCKET_IFNAME"] = "lo"
+    os.environ["NCCL_SOCKET_IFNAME"] = os.environ.get("BNET_TEST_IFNAME",
+                                                      "lo")
     os.environ["BNET_MIN_CHUNKSIZE"] = "8192"
     os.environ["BNET_NSTREAMS"] = "3"
 
@@ -81,7 +82,7 @@ def _sender(conn, result):
     p.close_send(scomm)
 
 
-def test_two_process_transfer():
+def _run_pair():
     ctx = mp.get_context("spawn")
     a, b = ctx.Pipe()
     res = ctx.Queue()
@@ -99,3 +100,25 @@ def test_two_process_transfer():
         hashlib.sha256(_payload(i, s)).hexdigest() for i, s in enumerate(SIZES)
     ]
     assert digests == expect
+
+
+def test_two_process_transfer():
+    _run_pair()
+
+
+def test_two_process_transfer_real_nic():
+    """Same, over the default (non-loopback) interface — exercises the NIC
+    addressing/bind path used inter-node."""
+    import pytest
+
+    try:
+        ifaces = os.listdir("/sys/class/net")
+    except OSError:
+        ifaces = []
+    if not [i for i in ifaces if i != "lo" and not i.startswith("docker")]:
+        pytest.skip("no non-loopback interface")
+    os.environ["BNET_TEST_IFNAME"] = "^docker,lo"
+    try:
+        _run_pair()
+    finally:
+        del os.environ["BNET_TEST_IFNAME"]
