@@ -54,6 +54,10 @@ def main():
                 n = max(size // 4, 1)
                 src = torch.randn(n, device="cuda")
                 dst = torch.zeros_like(src)
+                # ncclNet semantics: buffers must be ready at isend/irecv
+                # time (NCCL's proxy guarantees this before calling the
+                # plugin) — the producer kernels run async, so sync first
+                torch.cuda.synchronize()
                 nbytes = n * 4
                 smh = p.reg_mr(scomm, C.c_void_p(src.data_ptr()), nbytes,
                                0x2)

@@ -104,7 +104,7 @@ def bench_plugin_gpu_loopback():
         n_msgs = max(8, (1 << 30) // size // 4)
         src = torch.randn(size // 4, device="cuda")
         dsts = [torch.empty_like(src) for _ in range(DEPTH)]
-        torch.cuda.synchronize()
+        torch.cuda.synchronize()  # buffers ready before posting (ABI)
         smh = p.reg_mr(scomm, C.c_void_p(src.data_ptr()), size, 0x2)
         rmh = p.reg_mr(rcomm, C.c_void_p(dsts[0].data_ptr()), size, 0x2)
         t0 = time.perf_counter()

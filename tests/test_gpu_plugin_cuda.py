@@ -28,6 +28,8 @@ def gconn(plugin):
 
 
 def gpu_xfer(plugin, scomm, rcomm, src: torch.Tensor, dst: torch.Tensor):
+    # ncclNet contract: buffers are ready when isend/irecv is called
+    torch.cuda.synchronize()
     size = src.numel() * src.element_size()
     smh = plugin.reg_mr(scomm, C.c_void_p(src.data_ptr()), size,
                         NCCL_PTR_CUDA)
