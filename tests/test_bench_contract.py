@@ -1,0 +1,52 @@
+"""Driver-contract regression tests: bench.py must emit exactly one valid
+JSON line with the agreed keys, and the torchrun launch path must work."""
+
+import json
+import os
+import subprocess
+import sys
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def test_bench_json_contract():
+    res = subprocess.run(
+        [sys.executable, os.path.join(REPO, "bench.py"), "--model",
+         "resnet50", "--steps", "1", "--warmup", "0", "--batch", "2"],
+        capture_output=True, text=True, timeout=300, cwd=REPO,
+    )
+    assert res.returncode == 0, res.stderr[-2000:]
+    lines = [ln for ln in res.stdout.splitlines() if ln.startswith("{")]
+    assert len(lines) == 1, res.stdout
+    out = json.loads(lines[0])
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling",
+                "vs_baseline", "dtype", "data", "config"):
+        assert key in out, f"missing key {key}"
+    assert out["n_gpus"] == 1
+    assert out["steps"] == 1
+    assert out["unit"] == "img/sec"
+    assert out["higher_is_better"] is True
+    assert out["scaling"] == "weak"
+    assert out["data"] == "synthetic"
+    assert isinstance(out["value"], (int, float)) and out["value"] > 0
+    assert out["config"]["global_batch"] == 2
+    assert out["config"]["parallelism"] == "dp1"
+
+
+def test_allreduce_perf_torchrun_cpu():
+    env = dict(os.environ)
+    res = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29533",
+         os.path.join(REPO, "benchmarks", "allreduce_perf.py"),
+         "--max-bytes", "16384", "--iters", "2", "--warmup", "1"],
+        capture_output=True, text=True, timeout=300, cwd=REPO, env=env,
+    )
+    assert res.returncode == 0, res.stderr[-2000:]
+    summary = [ln for ln in res.stdout.splitlines()
+               if ln.startswith('{"bench"')]
+    assert summary, res.stdout
+    s = json.loads(summary[0])
+    assert s["world"] == 2 and s["backend"] == "gloo"
