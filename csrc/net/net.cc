@@ -336,7 +336,7 @@ ncclResult_t Net::isend(void* send_comm, void* data, int size, int tag,
   auto* c = (SendComm*)send_comm;
   if (c->error.load(std::memory_order_relaxed)) return ncclSystemError;
   SendRequest* r = &c->reqs[c->seq_next % NCCL_NET_MAX_REQUESTS];
-  if (r->state.load(std::memory_order_acquire) != REQ_FREE) {
+  if (ss_state(r->state_seq.load(std::memory_order_acquire)) != REQ_FREE) {
     *request = nullptr;  // slot busy — NCCL retries
     return ncclSuccess;
   }
@@ -369,20 +369,19 @@ ncclResult_t Net::isend(void* send_comm, void* data, int size, int tag,
     r->src = (const char*)data;
     r->avail.store((uint32_t)size, std::memory_order_relaxed);
   }
-  r->seq.store(c->seq_next, std::memory_order_relaxed);
-  r->state.store(REQ_ACTIVE, std::memory_order_release);
+  uint32_t seq = c->seq_next;
+  r->state_seq.store(pack_ss(seq, REQ_ACTIVE), std::memory_order_release);
   c->seq_next++;
   c->stats.isend_count.fetch_add(1, std::memory_order_relaxed);
   auto& T = Telemetry::get();
   T.isend_count.fetch_add(1, std::memory_order_relaxed);
   T.hist_add(T.isend_hist, (uint64_t)size);
-  r->span_slot = T.span_begin(0, (uint64_t)(uintptr_t)c, r->seq.load(),
+  r->span_slot = T.span_begin(0, (uint64_t)(uintptr_t)c, seq,
                               (uint32_t)size);
   int nchunks = r->total ? (int)((r->total + r->chunk - 1) / r->chunk) : 1;
   if (nchunks <= 1) {
     size_t nsmall = c->socks.size() < 2 ? c->socks.size() : 2;
-    Engine::get().kick_sock(
-        c->socks[r->seq.load(std::memory_order_relaxed) % nsmall]);
+    Engine::get().kick_sock(c->socks[seq % nsmall]);
   }
   else
     Engine::get().kick_comm(c, nchunks);
@@ -400,7 +399,7 @@ ncclResult_t Net::irecv(void* recv_comm, int n, void** data, int* sizes,
     return ncclInternalError;
   }
   RecvRequest* r = &c->reqs[c->post_next % NCCL_NET_MAX_REQUESTS];
-  if (r->state.load(std::memory_order_acquire) != REQ_FREE) {
+  if (ss_state(r->state_seq.load(std::memory_order_acquire)) != REQ_FREE) {
     *request = nullptr;
     return ncclSuccess;
   }
@@ -428,14 +427,14 @@ ncclResult_t Net::irecv(void* recv_comm, int n, void** data, int* sizes,
   } else {
     r->stage = nullptr;
   }
-  r->seq.store(c->post_next, std::memory_order_relaxed);
-  r->state.store(REQ_ACTIVE, std::memory_order_release);
+  uint32_t seq = c->post_next;
+  r->state_seq.store(pack_ss(seq, REQ_ACTIVE), std::memory_order_release);
   c->post_next++;
   c->stats.irecv_count.fetch_add(1, std::memory_order_relaxed);
   auto& T = Telemetry::get();
   T.irecv_count.fetch_add(1, std::memory_order_relaxed);
   T.hist_add(T.irecv_hist, (uint64_t)sizes[0]);
-  r->span_slot = T.span_begin(1, (uint64_t)(uintptr_t)c, r->seq.load(),
+  r->span_slot = T.span_begin(1, (uint64_t)(uintptr_t)c, seq,
                               (uint32_t)sizes[0]);
   Engine::get().kick_comm(c);  // wake sockets parked on this seq
   *request = tag_recv(r);
@@ -470,7 +469,9 @@ ncclResult_t Net::test(void* request, int* done, int* sizes) {
       T.bytes_sent.fetch_add(r->total, std::memory_order_relaxed);
       T.span_end(r->span_slot);
       if (r->stage) stage_release(c->stage_pool, r);
-      r->state.store(REQ_FREE, std::memory_order_release);
+      uint64_t ss = r->state_seq.load(std::memory_order_relaxed);
+      r->state_seq.store(pack_ss(ss_seq(ss), REQ_FREE),
+                         std::memory_order_release);
     }
     return ncclSuccess;
   }
@@ -489,7 +490,9 @@ ncclResult_t Net::test(void* request, int* done, int* sizes) {
                              std::memory_order_relaxed);
       T.span_end(r->span_slot);
       if (r->stage) stage_release(c->stage_pool, r);
-      r->state.store(REQ_FREE, std::memory_order_release);
+      uint64_t ss = r->state_seq.load(std::memory_order_relaxed);
+      r->state_seq.store(pack_ss(ss_seq(ss), REQ_FREE),
+                         std::memory_order_release);
     }
     return ncclSuccess;
   }

@@ -224,8 +224,9 @@ static SendRequest* claim_chunk(SendComm* c, int sock_idx, uint32_t* off,
   uint32_t newest = oldest + NCCL_NET_MAX_REQUESTS;
   for (uint32_t s = oldest; s != newest; s++) {
     SendRequest* r = &c->reqs[s % NCCL_NET_MAX_REQUESTS];
-    uint32_t st = r->state.load(std::memory_order_acquire);
-    uint32_t rseq = r->seq.load(std::memory_order_relaxed);
+    uint64_t ss = r->state_seq.load(std::memory_order_acquire);
+    uint32_t st = ss_state(ss);
+    uint32_t rseq = ss_seq(ss);
     if (st != REQ_ACTIVE || rseq != s) {
       if (rseq == s || (st == REQ_ACTIVE && rseq != s)) {
         // seq s already completed (slot FREE with our seq, or reused by a
@@ -286,8 +287,8 @@ void IoThread::progress_send(TcpSock* s) {
       }
       s->tx.active = true;
       s->tx.req = r;
-      s->tx.hdr = {r->seq.load(std::memory_order_relaxed), off, len,
-                   r->total};
+      s->tx.hdr = {ss_seq(r->state_seq.load(std::memory_order_relaxed)), off,
+                   len, r->total};
       s->tx.payload = r->src + off;
       s->tx.done = 0;
     }
@@ -359,7 +360,8 @@ void IoThread::progress_recv(TcpSock* s) {
           // EOF on a message boundary: benign iff no recv is pending
           bool pending = false;
           for (auto& r : c->reqs)
-            if (r.state.load(std::memory_order_acquire) == REQ_ACTIVE)
+            if (ss_state(r.state_seq.load(std::memory_order_acquire)) ==
+                REQ_ACTIVE)
               pending = true;
           if (!pending) {
             epoll_event ev{};
@@ -377,8 +379,8 @@ void IoThread::progress_recv(TcpSock* s) {
       // header complete → locate the posted request
       uint32_t seq = s->rx.hdr.seq;
       RecvRequest* r = &c->reqs[seq % NCCL_NET_MAX_REQUESTS];
-      if (r->state.load(std::memory_order_acquire) != REQ_ACTIVE ||
-          r->seq.load(std::memory_order_relaxed) != seq) {
+      if (r->state_seq.load(std::memory_order_acquire) !=
+          pack_ss(seq, REQ_ACTIVE)) {
         // not posted yet — park; irecv() will kick us
         if (!s->parked.exchange(true)) {
           epoll_event ev{};

@@ -88,12 +88,22 @@ static_assert(sizeof(ListenHandle) <= NCCL_NET_HANDLE_MAXSIZE, "handle size");
 
 enum ReqState : uint32_t { REQ_FREE = 0, REQ_ACTIVE = 1 };
 
+// state+seq packed into one atomic word so observers can never see a torn
+// (state, seq) pair.  Without this, a claim scan could read stale
+// state=FREE together with a freshly-written new seq during slot re-posting
+// and misclassify an in-flight post as "completed", advancing the send
+// frontier past a request that was never transmitted.
+inline uint64_t pack_ss(uint32_t seq, uint32_t state) {
+  return ((uint64_t)seq << 32) | state;
+}
+inline uint32_t ss_state(uint64_t ss) { return (uint32_t)ss; }
+inline uint32_t ss_seq(uint64_t ss) { return (uint32_t)(ss >> 32); }
+
 struct SendComm;
 struct RecvComm;
 
 struct SendRequest {
-  std::atomic<uint32_t> state{REQ_FREE};
-  std::atomic<uint32_t> seq{UINT32_MAX};  // UINT32_MAX = never used
+  std::atomic<uint64_t> state_seq{pack_ss(UINT32_MAX, REQ_FREE)};
   uint32_t total = 0;
   uint32_t chunk = 0;           // stripe chunk size chosen for this message
   const char* src = nullptr;    // host source (user buffer or staging bounce)
@@ -113,8 +123,7 @@ struct SendRequest {
 };
 
 struct RecvRequest {
-  std::atomic<uint32_t> state{REQ_FREE};
-  std::atomic<uint32_t> seq{UINT32_MAX};  // UINT32_MAX = never used
+  std::atomic<uint64_t> state_seq{pack_ss(UINT32_MAX, REQ_FREE)};
   char* dst = nullptr;     // user destination (host) — staging writes here
   uint32_t capacity = 0;   // posted buffer size (recv may be smaller)
   std::atomic<int64_t> total{-1};      // from first chunk header
