@@ -19,12 +19,22 @@ def t(prefix):
     return m[0] if m else None
 
 
+strings = {}
+st = t("rocpd_string")
+if st:
+    cols = [c[1] for c in cur.execute(f"PRAGMA table_info({st})")]
+    val = "string" if "string" in cols else cols[-1]
+    for i, v in cur.execute(f"SELECT id, {val} FROM {st}"):
+        strings[i] = v
+
 rows = []
 mc = t("rocpd_memory_copy")
 if mc:
-    for s, e, name in cur.execute(
-            f"SELECT start, end, name FROM {mc} ORDER BY start"):
-        rows.append((s, e, "copy:" + (name or "?")))
+    for s, e, nid, size, stream in cur.execute(
+            f"SELECT start, end, name_id, size, stream_id FROM {mc} "
+            "ORDER BY start"):
+        rows.append((s, e, f"copy:{strings.get(nid, nid)}:{size}B:"
+                           f"stream{stream}"))
 kd = t("rocpd_kernel_dispatch")
 ks = t("rocpd_info_kernel_symbol")
 if kd and ks:
