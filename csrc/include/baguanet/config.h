@@ -40,6 +40,10 @@ struct Config {
   int stage_kernel = 0;
   // Listen backlog (reference: 16384, nthread:101).
   int backlog = 16384;
+  // IO engine: "EPOLL" (default) or "URING" (io_uring; falls back to
+  // epoll when unavailable).  The reference's BAGUA_NET_IMPLEMENT
+  // BASIC/TOKIO selector, rebuilt as readiness- vs completion-based IO.
+  std::string implement = "EPOLL";
   // Metrics dump file ("" = disabled); written at process exit.
   std::string metrics_file;
   // Chrome-trace span dump file ("" = disabled).

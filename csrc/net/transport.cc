@@ -212,8 +212,8 @@ void IoThread::progress(TcpSock* s) {
 // just bounces the slot cache lines across 4 IO threads and halves the
 // small-message rate (measured on the MI355X box: 3.8 vs 11.4 GB/s at
 // 64 KiB).  Multi-chunk messages keep fully dynamic assignment.
-static SendRequest* claim_chunk(SendComm* c, int sock_idx, uint32_t* off,
-                                uint32_t* len) {
+SendRequest* claim_chunk(SendComm* c, int sock_idx, uint32_t* off,
+                         uint32_t* len) {
   int nsocks = (int)c->socks.size();
   // Single-chunk messages fan out over at most 2 sockets: measured on the
   // MI355X box, 64 KiB message rate is ~11 GB/s at 1-2 sockets but drops
@@ -377,10 +377,9 @@ void IoThread::progress_recv(TcpSock* s) {
         return;
       }
       // header complete → locate the posted request
-      uint32_t seq = s->rx.hdr.seq;
-      RecvRequest* r = &c->reqs[seq % NCCL_NET_MAX_REQUESTS];
-      if (r->state_seq.load(std::memory_order_acquire) !=
-          pack_ss(seq, REQ_ACTIVE)) {
+      int hr = process_recv_header(s);
+      if (hr < 0) return;  // protocol error (comm error set)
+      if (hr == 1) {
         // not posted yet — park; irecv() will kick us
         if (!s->parked.exchange(true)) {
           epoll_event ev{};
@@ -396,28 +395,9 @@ void IoThread::progress_recv(TcpSock* s) {
         ev.data.ptr = s;
         epoll_ctl(epfd_, EPOLL_CTL_MOD, s->fd, &ev);
       }
-      int64_t expect = -1;
-      r->total.compare_exchange_strong(expect, (int64_t)s->rx.hdr.total);
-      if ((int64_t)s->rx.hdr.total != r->total.load()) {
-        BNET_WARN("bnet: inconsistent total in chunk headers (%u vs %ld)",
-                  s->rx.hdr.total, (long)r->total.load());
-        c->error.store(EPROTO);
-        return;
-      }
-      if (s->rx.hdr.total > r->capacity) {
-        BNET_WARN("bnet: message (%u B) exceeds posted buffer (%u B)",
-                  s->rx.hdr.total, r->capacity);
-        c->error.store(EMSGSIZE);
-        return;
-      }
-      s->rx.req = r;
-      char* base = stage_recv_base(r);  // bounce for CUDA dst, else dst
-      s->rx.target = base + s->rx.hdr.offset;
-      s->rx.remaining = s->rx.hdr.len;
-      s->rx.in_payload = true;
       if (s->rx.hdr.len == 0) {
         // empty chunk (only for zero-byte messages)
-        finish_chunk(s);
+        finish_rx_chunk(s);
         continue;
       }
     }
@@ -435,12 +415,42 @@ void IoThread::progress_recv(TcpSock* s) {
                                                : strerror(errno));
       return;
     }
-    finish_chunk(s);
+    finish_rx_chunk(s);
   }
 }
 
+// rx.hdr is complete: locate/validate the posted request (see transport.h).
+int process_recv_header(TcpSock* s) {
+  RecvComm* c = s->rcomm;
+  uint32_t seq = s->rx.hdr.seq;
+  RecvRequest* r = &c->reqs[seq % NCCL_NET_MAX_REQUESTS];
+  if (r->state_seq.load(std::memory_order_acquire) !=
+      pack_ss(seq, REQ_ACTIVE))
+    return 1;  // not posted yet
+  int64_t expect = -1;
+  r->total.compare_exchange_strong(expect, (int64_t)s->rx.hdr.total);
+  if ((int64_t)s->rx.hdr.total != r->total.load()) {
+    BNET_WARN("bnet: inconsistent total in chunk headers (%u vs %ld)",
+              s->rx.hdr.total, (long)r->total.load());
+    c->error.store(EPROTO);
+    return -1;
+  }
+  if (s->rx.hdr.total > r->capacity) {
+    BNET_WARN("bnet: message (%u B) exceeds posted buffer (%u B)",
+              s->rx.hdr.total, r->capacity);
+    c->error.store(EMSGSIZE);
+    return -1;
+  }
+  s->rx.req = r;
+  char* base = stage_recv_base(r);  // bounce for CUDA dst, else dst
+  s->rx.target = base + s->rx.hdr.offset;
+  s->rx.remaining = s->rx.hdr.len;
+  s->rx.in_payload = true;
+  return 0;
+}
+
 // Called with a fully-received chunk in s->rx.
-void IoThread::finish_chunk(TcpSock* s) {
+void finish_rx_chunk(TcpSock* s) {
   RecvComm* c = s->rcomm;
   RecvRequest* r = s->rx.req;
   ChunkHdr h = s->rx.hdr;
@@ -461,12 +471,26 @@ void IoThread::finish_chunk(TcpSock* s) {
 
 Engine::Engine() {
   int n = Config::get().io_threads;
-  threads_ = std::vector<IoThread>(n);
-  for (int i = 0; i < n; i++) threads_[i].start(i);
+  bool want_uring = Config::get().implement == "URING";
+  if (want_uring && !uring_available()) {
+    BNET_WARN("BNET_IMPLEMENT=URING but io_uring is unavailable here "
+              "(seccomp?) — falling back to the epoll engine");
+    want_uring = false;
+  }
+  impl_ = want_uring ? "URING" : "EPOLL";
+  threads_.reserve(n);
+  for (int i = 0; i < n; i++) {
+    if (want_uring)
+      threads_.emplace_back(make_uring_thread());
+    else
+      threads_.emplace_back(new IoThread());
+    threads_.back()->start(i);
+  }
+  BNET_INFO("baguanet engine: %s, %d IO thread(s)", impl_, n);
 }
 
 Engine::~Engine() {
-  for (auto& t : threads_) t.stop();
+  for (auto& t : threads_) t->stop();
 }
 
 Engine& Engine::get() {
@@ -480,11 +504,11 @@ int Engine::assign() {
 
 void Engine::register_sock(TcpSock* s) {
   s->io_thread = assign();
-  threads_[s->io_thread].add_sock(s);
+  threads_[s->io_thread]->add_sock(s);
 }
 
 void Engine::unregister_sock_sync(TcpSock* s) {
-  if (s->io_thread >= 0) threads_[s->io_thread].remove_sock_sync(s);
+  if (s->io_thread >= 0) threads_[s->io_thread]->remove_sock_sync(s);
 }
 
 void Engine::kick_comm(SendComm* c, int max_socks) {
@@ -495,20 +519,20 @@ void Engine::kick_comm(SendComm* c, int max_socks) {
   for (TcpSock* s : c->socks) {
     if (n <= 0) break;
     if (!s->tx.active && !s->want_epollout) {
-      threads_[s->io_thread].kick(s);
+      threads_[s->io_thread]->kick(s);
       n--;
     }
   }
 }
 
 void Engine::kick_sock(TcpSock* s) {
-  if (!s->tx.active && !s->want_epollout) threads_[s->io_thread].kick(s);
+  if (!s->tx.active && !s->want_epollout) threads_[s->io_thread]->kick(s);
 }
 
 void Engine::kick_comm(RecvComm* c) {
   for (TcpSock* s : c->socks)
     if (s->parked.load(std::memory_order_relaxed))
-      threads_[s->io_thread].kick(s);
+      threads_[s->io_thread]->kick(s);
 }
 
 }  // namespace baguanet

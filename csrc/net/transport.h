@@ -30,6 +30,7 @@
 
 #include <netinet/in.h>
 #include <sys/socket.h>
+#include <sys/uio.h>
 
 #include <atomic>
 #include <condition_variable>
@@ -147,6 +148,13 @@ struct TcpSock {
   int idx = 0;  // position within its comm's socket list
   bool is_recv = false;
   bool want_epollout = false;
+  // io_uring engine per-socket state (unused by the epoll engine)
+  struct {
+    uint8_t op = 0;  // 0 none, 1 send(writev), 2 recv
+    bool closing = false;  // removal in progress: no resubmission
+    struct iovec iov[2];
+    uint32_t want = 0;  // bytes requested by the outstanding recv
+  } ur;
   std::atomic<bool> parked{false};  // recv: waiting for a not-yet-posted seq
   SendComm* scomm = nullptr;
   RecvComm* rcomm = nullptr;
@@ -238,13 +246,33 @@ struct ListenComm {
 
 // -------------------------------------------------------------- engine ----
 
-class IoThread {
+// Shared transport state-machine helpers (used by both engines).
+SendRequest* claim_chunk(SendComm* c, int sock_idx, uint32_t* off,
+                         uint32_t* len);
+// rx.hdr is complete: locate/validate the posted request and prime
+// rx.target/remaining.  Returns 0 = proceed, 1 = park (request not yet
+// posted), -1 = protocol error (comm error set).
+int process_recv_header(TcpSock* s);
+// A fully-received chunk in s->rx: account + staging hand-off.
+void finish_rx_chunk(TcpSock* s);
+
+class IIoThread {
  public:
-  void start(int idx);
-  void stop();
-  void add_sock(TcpSock* s);                 // thread-safe
-  void remove_sock_sync(TcpSock* s);         // blocks until removed
-  void kick(TcpSock* s);                     // re-run progress for s
+  virtual ~IIoThread() = default;
+  virtual void start(int idx) = 0;
+  virtual void stop() = 0;
+  virtual void add_sock(TcpSock* s) = 0;          // thread-safe
+  virtual void remove_sock_sync(TcpSock* s) = 0;  // blocks until removed
+  virtual void kick(TcpSock* s) = 0;              // re-run progress for s
+};
+
+class IoThread : public IIoThread {
+ public:
+  void start(int idx) override;
+  void stop() override;
+  void add_sock(TcpSock* s) override;
+  void remove_sock_sync(TcpSock* s) override;
+  void kick(TcpSock* s) override;
   void kick_all();
 
  private:
@@ -252,7 +280,6 @@ class IoThread {
   void progress(TcpSock* s);
   void progress_send(TcpSock* s);
   void progress_recv(TcpSock* s);
-  void finish_chunk(TcpSock* s);
   void set_epollout(TcpSock* s, bool on);
   void handle_tasks();
 
@@ -271,10 +298,16 @@ class IoThread {
   std::vector<TcpSock*> socks_;  // owned set (IO thread only)
 };
 
+// io_uring-based engine thread (uring_engine.cc); selected with
+// BNET_IMPLEMENT=URING, falls back to epoll when io_uring is unavailable
+// (e.g. seccomp-filtered containers).
+IIoThread* make_uring_thread();
+bool uring_available();
+
 class Engine {
  public:
   static Engine& get();
-  IoThread& thread(int idx) { return threads_[idx]; }
+  IIoThread& thread(int idx) { return *threads_[idx]; }
   int assign();  // round-robin IO thread index
   int nthreads() const { return (int)threads_.size(); }
   void register_sock(TcpSock* s);
@@ -282,12 +315,14 @@ class Engine {
   void kick_comm(SendComm* c, int max_socks = -1);
   void kick_comm(RecvComm* c);
   void kick_sock(TcpSock* s);
+  const char* impl() const { return impl_; }
 
  private:
   Engine();
   ~Engine();
-  std::vector<IoThread> threads_;
+  std::vector<std::unique_ptr<IIoThread>> threads_;
   std::atomic<uint32_t> rr_{0};
+  const char* impl_ = "EPOLL";
 };
 
 // ------------------------------------------------------------- net API ----

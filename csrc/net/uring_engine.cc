@@ -1,0 +1,517 @@
+// uring_engine.cc — io_uring completion-based IO engine (BNET_IMPLEMENT=URING).
+//
+// The second engine, paralleling the reference's BASIC/TOKIO backend pair
+//  the modern-Linux way: the epoll engine is
+// readiness-based (wait → syscall read/writev per span), this one is
+// completion-based — SQEs for WRITEV/RECV are batched into one
+// io_uring_enter and the kernel reports finished transfers through the CQ
+// ring.  Raw syscalls + mmap'd rings (no liburing in the image).
+//
+// Semantics are identical to the epoll engine: same chunk claiming
+// (claim_chunk), same header processing (process_recv_header), same
+// completion accounting (finish_rx_chunk), same parking rules.  At most
+// one SQE is in flight per socket; its completion immediately claims and
+// submits the next span, so the kernel always has work queued.
+
+#include <errno.h>
+#include <linux/io_uring.h>
+#include <string.h>
+#include <sys/eventfd.h>
+#include <sys/mman.h>
+#include <sys/syscall.h>
+#include <unistd.h>
+
+#include <algorithm>
+#include <atomic>
+#include <condition_variable>
+#include <mutex>
+#include <thread>
+#include <vector>
+
+#include "baguanet/config.h"
+#include "baguanet/log.h"
+#include "staging.h"
+#include "telemetry.h"
+#include "transport.h"
+
+namespace baguanet {
+
+namespace {
+
+int sys_uring_setup(unsigned entries, io_uring_params* p) {
+  return (int)syscall(__NR_io_uring_setup, entries, p);
+}
+int sys_uring_enter(int fd, unsigned to_submit, unsigned min_complete,
+                    unsigned flags) {
+  return (int)syscall(__NR_io_uring_enter, fd, to_submit, min_complete,
+                      flags, nullptr, 0);
+}
+
+class UringRing {
+ public:
+  bool init(unsigned entries) {
+    io_uring_params p{};
+    fd_ = sys_uring_setup(entries, &p);
+    if (fd_ < 0) return false;
+    if (!(p.features & IORING_FEAT_SINGLE_MMAP)) {
+      close(fd_);
+      fd_ = -1;
+      return false;  // pre-5.4 kernels — not our targets
+    }
+    sq_sz_ = p.sq_off.array + p.sq_entries * sizeof(unsigned);
+    size_t cq_sz = p.cq_off.cqes + p.cq_entries * sizeof(io_uring_cqe);
+    if (cq_sz > sq_sz_) sq_sz_ = cq_sz;
+    ring_ = mmap(nullptr, sq_sz_, PROT_READ | PROT_WRITE,
+                 MAP_SHARED | MAP_POPULATE, fd_, IORING_OFF_SQ_RING);
+    sqes_sz_ = p.sq_entries * sizeof(io_uring_sqe);
+    sqes_ = (io_uring_sqe*)mmap(nullptr, sqes_sz_, PROT_READ | PROT_WRITE,
+                                MAP_SHARED | MAP_POPULATE, fd_,
+                                IORING_OFF_SQES);
+    if (ring_ == MAP_FAILED || sqes_ == MAP_FAILED) {
+      close(fd_);
+      fd_ = -1;
+      return false;
+    }
+    char* r = (char*)ring_;
+    sq_head_ = (std::atomic<unsigned>*)(r + p.sq_off.head);
+    sq_tail_ = (std::atomic<unsigned>*)(r + p.sq_off.tail);
+    sq_mask_ = *(unsigned*)(r + p.sq_off.ring_mask);
+    sq_array_ = (unsigned*)(r + p.sq_off.array);
+    cq_head_ = (std::atomic<unsigned>*)(r + p.cq_off.head);
+    cq_tail_ = (std::atomic<unsigned>*)(r + p.cq_off.tail);
+    cq_mask_ = *(unsigned*)(r + p.cq_off.ring_mask);
+    cqes_ = (io_uring_cqe*)(r + p.cq_off.cqes);
+    entries_ = p.sq_entries;
+    return true;
+  }
+  ~UringRing() {
+    if (ring_ && ring_ != MAP_FAILED) munmap(ring_, sq_sz_);
+    if (sqes_ && sqes_ != (io_uring_sqe*)MAP_FAILED) munmap(sqes_, sqes_sz_);
+    if (fd_ >= 0) close(fd_);
+  }
+
+  io_uring_sqe* get_sqe() {
+    unsigned tail = sq_tail_->load(std::memory_order_relaxed);
+    unsigned head = sq_head_->load(std::memory_order_acquire);
+    if (tail - head >= entries_) return nullptr;  // ring full
+    io_uring_sqe* sqe = &sqes_[tail & sq_mask_];
+    memset(sqe, 0, sizeof(*sqe));
+    sq_array_[tail & sq_mask_] = tail & sq_mask_;
+    pending_tail_ = tail + 1;
+    return sqe;
+  }
+  void advance_tail() {
+    sq_tail_->store(pending_tail_, std::memory_order_release);
+    to_submit_++;
+  }
+  // submit queued SQEs; wait for >= min_complete completions
+  int enter(unsigned min_complete) {
+    unsigned n = to_submit_;
+    to_submit_ = 0;
+    int rc = sys_uring_enter(fd_, n, min_complete,
+                             min_complete ? IORING_ENTER_GETEVENTS : 0);
+    return rc;
+  }
+  io_uring_cqe* peek() {
+    unsigned head = cq_head_->load(std::memory_order_relaxed);
+    if (head == cq_tail_->load(std::memory_order_acquire)) return nullptr;
+    return &cqes_[head & cq_mask_];
+  }
+  void seen() {
+    cq_head_->fetch_add(1, std::memory_order_release);
+  }
+
+ private:
+  int fd_ = -1;
+  void* ring_ = nullptr;
+  io_uring_sqe* sqes_ = nullptr;
+  size_t sq_sz_ = 0, sqes_sz_ = 0;
+  std::atomic<unsigned>* sq_head_ = nullptr;
+  std::atomic<unsigned>* sq_tail_ = nullptr;
+  unsigned sq_mask_ = 0;
+  unsigned* sq_array_ = nullptr;
+  std::atomic<unsigned>* cq_head_ = nullptr;
+  std::atomic<unsigned>* cq_tail_ = nullptr;
+  unsigned cq_mask_ = 0;
+  io_uring_cqe* cqes_ = nullptr;
+  unsigned entries_ = 0;
+  unsigned pending_tail_ = 0;
+  unsigned to_submit_ = 0;
+};
+
+constexpr uint64_t kUdEventfd = 1;  // user_data for the eventfd read op
+
+class UringIoThread : public IIoThread {
+ public:
+  void start(int idx) override {
+    idx_ = idx;
+    evfd_ = eventfd(0, EFD_CLOEXEC);  // blocking read op via the ring
+    if (!ring_.init(256)) {
+      // engine factory guarantees availability; treat as fatal for thread
+      BNET_WARN("io_uring init failed mid-run: %s", strerror(errno));
+      return;
+    }
+    thr_ = std::thread([this] { run(); });
+  }
+  void stop() override {
+    if (!thr_.joinable()) return;
+    stop_.store(true);
+    uint64_t one = 1;
+    (void)!write(evfd_, &one, sizeof(one));
+    thr_.join();
+    close(evfd_);
+  }
+  void add_sock(TcpSock* s) override {
+    enqueue({Task::ADD, s, nullptr, nullptr});
+  }
+  void remove_sock_sync(TcpSock* s) override {
+    std::mutex mu;
+    std::condition_variable cv;
+    bool done = false;
+    enqueue({Task::REMOVE, s, &cv, &done});
+    std::unique_lock<std::mutex> lk(mu);
+    while (true) {
+      {
+        std::lock_guard<std::mutex> l2(task_mu_);
+        if (done) break;
+      }
+      cv.wait_for(lk, std::chrono::milliseconds(1));
+    }
+  }
+  void kick(TcpSock* s) override { enqueue({Task::KICK, s, nullptr, nullptr}); }
+
+ private:
+  struct Task {
+    enum { ADD, REMOVE, KICK } kind;
+    TcpSock* s;
+    std::condition_variable* cv;
+    bool* flag;
+  };
+
+  void enqueue(Task t) {
+    {
+      std::lock_guard<std::mutex> lk(task_mu_);
+      tasks_.push_back(t);
+    }
+    uint64_t one = 1;
+    (void)!write(evfd_, &one, sizeof(one));
+  }
+
+  void arm_eventfd() {
+    io_uring_sqe* sqe = ring_.get_sqe();
+    if (!sqe) return;  // ring full — re-armed on next drain
+    sqe->opcode = IORING_OP_READ;
+    sqe->fd = evfd_;
+    sqe->addr = (uint64_t)(uintptr_t)&ev_buf_;
+    sqe->len = sizeof(ev_buf_);
+    sqe->user_data = kUdEventfd;
+    ring_.advance_tail();
+    evfd_armed_ = true;
+  }
+
+  // ---- send path -------------------------------------------------------
+  void submit_send(TcpSock* s) {
+    SendComm* c = s->scomm;
+    if (s->ur.op || s->ur.closing ||
+        c->error.load(std::memory_order_relaxed))
+      return;
+    if (!s->tx.active) {
+      uint32_t off = 0, len = 0;
+      SendRequest* r = claim_chunk(c, s->idx, &off, &len);
+      if (!r) return;  // idle until next kick
+      s->tx.active = true;
+      s->tx.req = r;
+      s->tx.hdr = {ss_seq(r->state_seq.load(std::memory_order_relaxed)), off,
+                   len, r->total};
+      s->tx.payload = r->src + off;
+      s->tx.done = 0;
+    }
+    uint32_t hdr_left =
+        s->tx.done < sizeof(ChunkHdr) ? sizeof(ChunkHdr) - s->tx.done : 0;
+    uint32_t pay_done =
+        s->tx.done >= sizeof(ChunkHdr) ? s->tx.done - 16 : 0;
+    int iovn = 0;
+    if (hdr_left)
+      s->ur.iov[iovn++] = {(char*)&s->tx.hdr + (16 - hdr_left), hdr_left};
+    if (s->tx.hdr.len - pay_done)
+      s->ur.iov[iovn++] = {(void*)(s->tx.payload + pay_done),
+                           s->tx.hdr.len - pay_done};
+    if (iovn == 0) {  // nothing left (shouldn't happen)
+      complete_send_chunk(s);
+      return;
+    }
+    io_uring_sqe* sqe = ring_.get_sqe();
+    if (!sqe) return;  // ring full; retried after next drain
+    sqe->opcode = IORING_OP_WRITEV;
+    sqe->fd = s->fd;
+    sqe->addr = (uint64_t)(uintptr_t)s->ur.iov;
+    sqe->len = iovn;
+    sqe->user_data = (uint64_t)(uintptr_t)s;
+    ring_.advance_tail();
+    s->ur.op = 1;
+  }
+
+  void complete_send_chunk(TcpSock* s) {
+    SendComm* c = s->scomm;
+    SendRequest* r = s->tx.req;
+    uint32_t len = s->tx.hdr.len;
+    s->tx.active = false;
+    s->tx.req = nullptr;
+    if (r->total == 0) {
+      r->hdr_sent.store(true, std::memory_order_release);
+    } else {
+      c->stats.bytes_sent.fetch_add(len, std::memory_order_relaxed);
+      r->sent.fetch_add(len, std::memory_order_acq_rel);
+    }
+  }
+
+  void on_send_cqe(TcpSock* s, int res) {
+    s->ur.op = 0;
+    if (res <= 0) {
+      if (res == -ECANCELED) return;  // teardown cancel
+      if (res == -EAGAIN || res == -EINTR) {
+        submit_send(s);  // retry
+        return;
+      }
+      s->scomm->error.store(res ? -res : EPIPE);
+      BNET_WARN("bnet(uring) send error: %s", strerror(-res));
+      return;
+    }
+    s->tx.done += (uint32_t)res;
+    if (s->tx.done == 16 + s->tx.hdr.len) complete_send_chunk(s);
+    submit_send(s);  // next span / next chunk
+  }
+
+  // ---- recv path -------------------------------------------------------
+  void submit_recv(TcpSock* s) {
+    RecvComm* c = s->rcomm;
+    if (s->ur.op || s->ur.closing ||
+        c->error.load(std::memory_order_relaxed))
+      return;
+    if (s->parked.load(std::memory_order_relaxed)) return;
+    char* buf;
+    uint32_t want;
+    if (!s->rx.in_payload) {
+      buf = (char*)&s->rx.hdr + s->rx.hdr_got;
+      want = sizeof(ChunkHdr) - s->rx.hdr_got;
+    } else {
+      buf = s->rx.target;
+      want = s->rx.remaining;
+    }
+    io_uring_sqe* sqe = ring_.get_sqe();
+    if (!sqe) return;
+    sqe->opcode = IORING_OP_RECV;
+    sqe->fd = s->fd;
+    sqe->addr = (uint64_t)(uintptr_t)buf;
+    sqe->len = want;
+    sqe->user_data = (uint64_t)(uintptr_t)s;
+    ring_.advance_tail();
+    s->ur.op = 2;
+    s->ur.want = want;
+  }
+
+  void on_recv_cqe(TcpSock* s, int res) {
+    RecvComm* c = s->rcomm;
+    s->ur.op = 0;
+    if (res < 0) {
+      if (res == -ECANCELED) return;  // teardown cancel
+      if (res == -EAGAIN || res == -EINTR) {
+        submit_recv(s);
+        return;
+      }
+      c->error.store(-res);
+      BNET_WARN("bnet(uring) recv error: %s", strerror(-res));
+      return;
+    }
+    if (res == 0) {  // EOF
+      if (!s->rx.in_payload && s->rx.hdr_got == 0) {
+        bool pending = false;
+        for (auto& r : c->reqs)
+          if (ss_state(r.state_seq.load(std::memory_order_acquire)) ==
+              REQ_ACTIVE)
+            pending = true;
+        if (!pending) return;  // orderly shutdown: just stop re-arming
+      }
+      c->error.store(ECONNRESET);
+      BNET_WARN("bnet(uring) recv eof mid-protocol");
+      return;
+    }
+    if (!s->rx.in_payload) {
+      s->rx.hdr_got += (uint32_t)res;
+      if (s->rx.hdr_got == sizeof(ChunkHdr)) {
+        int hr = process_recv_header(s);
+        if (hr < 0) return;
+        if (hr == 1) {
+          s->parked.store(true);  // irecv() kick resumes us
+          return;
+        }
+        if (s->rx.hdr.len == 0) finish_rx_chunk(s);
+      }
+    } else {
+      s->rx.target += res;
+      s->rx.remaining -= (uint32_t)res;
+      if (s->rx.remaining == 0) finish_rx_chunk(s);
+    }
+    submit_recv(s);
+  }
+
+  void resume_parked(TcpSock* s) {
+    if (!s->parked.load(std::memory_order_relaxed)) return;
+    // header already buffered in rx.hdr — retry matching
+    int hr = process_recv_header(s);
+    if (hr < 0) return;
+    if (hr == 1) return;  // still unposted
+    s->parked.store(false);
+    if (s->rx.hdr.len == 0) finish_rx_chunk(s);
+    submit_recv(s);
+  }
+
+  void handle_tasks() {
+    std::vector<Task> batch;
+    {
+      std::lock_guard<std::mutex> lk(task_mu_);
+      batch.swap(tasks_);
+    }
+    for (auto& t : batch) {
+      switch (t.kind) {
+        case Task::ADD:
+          socks_.push_back(t.s);
+          if (t.s->is_recv)
+            submit_recv(t.s);
+          else
+            submit_send(t.s);
+          break;
+        case Task::REMOVE: {
+          TcpSock* s = t.s;
+          s->ur.closing = true;  // block resubmission from completions
+          // cancel any outstanding op on this fd, then reap its CQE
+          if (s->ur.op) {
+            io_uring_sqe* sqe = ring_.get_sqe();
+            if (sqe) {
+              sqe->opcode = IORING_OP_ASYNC_CANCEL;
+              sqe->addr = (uint64_t)(uintptr_t)s;
+              sqe->user_data = (uint64_t)(uintptr_t)s | 2;  // cancel marker
+              ring_.advance_tail();
+              ring_.enter(0);
+            }
+            // drain until the op's CQE arrives
+            int spins = 0;
+            while (s->ur.op && spins++ < 1000000) {
+              drain_cqes();
+              if (s->ur.op) ring_.enter(1);
+            }
+          }
+          close(s->fd);
+          s->fd = -1;
+          socks_.erase(std::remove(socks_.begin(), socks_.end(), s),
+                       socks_.end());
+          if (s->scomm) s->scomm->live_socks.fetch_sub(1);
+          if (s->rcomm) s->rcomm->live_socks.fetch_sub(1);
+          {
+            std::lock_guard<std::mutex> lk(task_mu_);
+            *t.flag = true;
+          }
+          t.cv->notify_all();
+          break;
+        }
+        case Task::KICK:
+          if (std::find(socks_.begin(), socks_.end(), t.s) != socks_.end()) {
+            if (t.s->is_recv)
+              resume_parked(t.s);
+            else
+              submit_send(t.s);
+          }
+          break;
+      }
+    }
+  }
+
+  void drain_cqes() {
+    io_uring_cqe* cqe;
+    while ((cqe = ring_.peek())) {
+      uint64_t ud = cqe->user_data;
+      int res = cqe->res;
+      ring_.seen();
+      if (ud == kUdEventfd) {
+        evfd_armed_ = false;
+        continue;
+      }
+      if (ud & 2) continue;  // ASYNC_CANCEL result — ignore
+      TcpSock* s = (TcpSock*)(uintptr_t)ud;
+      if (s->fd < 0) {
+        s->ur.op = 0;  // completion for a socket being removed
+        continue;
+      }
+      if (s->ur.op == 1)
+        on_send_cqe(s, res);
+      else if (s->ur.op == 2)
+        on_recv_cqe(s, res);
+    }
+  }
+
+  void run() {
+    char tname[16];
+    snprintf(tname, sizeof(tname), "bnet-ur%d", idx_);
+    pthread_setname_np(pthread_self(), tname);
+    arm_eventfd();
+    ring_.enter(0);
+    uint64_t last_active_ns = 0;
+    while (!stop_.load(std::memory_order_relaxed)) {
+      bool staging_busy = false;
+      for (TcpSock* s : socks_) {
+        if (s->scomm && s->scomm->stage_pool &&
+            stage_pending(s->scomm->stage_pool)) {
+          stage_poll(s->scomm->stage_pool);
+          staging_busy = true;
+        }
+      }
+      uint64_t now = now_ns();
+      bool spin = staging_busy || (now - last_active_ns < 200'000);
+      if (!evfd_armed_) arm_eventfd();
+      int rc = ring_.enter(spin ? 0 : 1);
+      (void)rc;
+      io_uring_cqe* had = ring_.peek();
+      if (had) last_active_ns = now;
+      drain_cqes();
+      handle_tasks();
+      if (spin || had) {
+        // jobs/watermarks may have advanced: retry idle senders
+        for (TcpSock* s : socks_)
+          if (!s->is_recv && !s->ur.op && !s->tx.active) submit_send(s);
+        ring_.enter(0);
+      }
+    }
+  }
+
+  int idx_ = 0;
+  int evfd_ = -1;
+  uint64_t ev_buf_ = 0;
+  bool evfd_armed_ = false;
+  UringRing ring_;
+  std::thread thr_;
+  std::atomic<bool> stop_{false};
+  std::mutex task_mu_;
+  std::vector<Task> tasks_;
+  std::vector<TcpSock*> socks_;
+};
+
+}  // namespace
+
+bool uring_available() {
+  static int avail = -1;
+  if (avail < 0) {
+    io_uring_params p{};
+    int fd = sys_uring_setup(4, &p);
+    if (fd >= 0) {
+      avail = (p.features & IORING_FEAT_SINGLE_MMAP) ? 1 : 0;
+      close(fd);
+    } else {
+      avail = 0;
+    }
+  }
+  return avail == 1;
+}
+
+IIoThread* make_uring_thread() { return new UringIoThread(); }
+
+}  // namespace baguanet
