@@ -513,26 +513,29 @@ void Engine::unregister_sock_sync(TcpSock* s) {
 
 void Engine::kick_comm(SendComm* c, int max_socks) {
   // Kicking more sockets than the message has chunks just burns eventfd
-  // wakeups (hurts small-message rates); idle sockets beyond the first
-  // chunk-count workers contribute nothing.
+  // wakeups (hurts small-message rates).  Kicks are UNCONDITIONAL for the
+  // chosen sockets: skipping an apparently-busy socket races with its
+  // transition to idle (it may have scanned for chunks before the new
+  // request was published and never look again) — a missed wakeup hangs
+  // the message.
   int n = max_socks < 0 ? (int)c->socks.size() : max_socks;
   for (TcpSock* s : c->socks) {
-    if (n <= 0) break;
-    if (!s->tx.active && !s->want_epollout) {
-      threads_[s->io_thread]->kick(s);
-      n--;
-    }
+    if (n-- <= 0) break;
+    threads_[s->io_thread]->kick(s);
   }
 }
 
 void Engine::kick_sock(TcpSock* s) {
-  if (!s->tx.active && !s->want_epollout) threads_[s->io_thread]->kick(s);
+  threads_[s->io_thread]->kick(s);  // unconditional — see kick_comm
 }
 
 void Engine::kick_comm(RecvComm* c) {
-  for (TcpSock* s : c->socks)
-    if (s->parked.load(std::memory_order_relaxed))
-      threads_[s->io_thread]->kick(s);
+  // UNCONDITIONAL: testing s->parked here races with the IO thread that
+  // is just deciding to park (it checked the slot before this irecv was
+  // published, we check parked before it sets the flag — both miss, and
+  // the socket sleeps forever with EPOLLIN disarmed).  A kick to an
+  // unparked socket is a cheap no-op read().
+  for (TcpSock* s : c->socks) threads_[s->io_thread]->kick(s);
 }
 
 }  // namespace baguanet

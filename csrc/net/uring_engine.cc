@@ -210,34 +210,53 @@ class UringIoThread : public IIoThread {
   }
 
   // ---- send path -------------------------------------------------------
+  // Claim up to kUrBatch chunks and submit them as ONE ordered WRITEV —
+  // multiple independent ops per socket would complete concurrently and
+  // interleave bytes, and a single chunk per op starves the kernel
+  // between completion and resubmission (measured 24 vs 47 GB/s against
+  // the epoll engine before batching).
   void submit_send(TcpSock* s) {
     SendComm* c = s->scomm;
     if (s->ur.op || s->ur.closing ||
         c->error.load(std::memory_order_relaxed))
       return;
-    if (!s->tx.active) {
-      uint32_t off = 0, len = 0;
-      SendRequest* r = claim_chunk(c, s->idx, &off, &len);
-      if (!r) return;  // idle until next kick
-      s->tx.active = true;
-      s->tx.req = r;
-      s->tx.hdr = {ss_seq(r->state_seq.load(std::memory_order_relaxed)), off,
-                   len, r->total};
-      s->tx.payload = r->src + off;
-      s->tx.done = 0;
+    if (s->ur.nchunks == 0) {
+      s->ur.batch_bytes = 0;
+      s->ur.done = 0;
+      while (s->ur.nchunks < TcpSock::kUrBatch) {
+        uint32_t off = 0, len = 0;
+        SendRequest* r = claim_chunk(c, s->idx, &off, &len);
+        if (!r) break;
+        int i = s->ur.nchunks++;
+        s->ur.reqs[i] = r;
+        s->ur.hdrs[i] = {ss_seq(r->state_seq.load(std::memory_order_relaxed)),
+                         off, len, r->total};
+        s->ur.payloads[i] = r->src + off;
+        s->ur.batch_bytes += 16 + len;
+      }
+      if (s->ur.nchunks == 0) return;  // idle until next kick
     }
-    uint32_t hdr_left =
-        s->tx.done < sizeof(ChunkHdr) ? sizeof(ChunkHdr) - s->tx.done : 0;
-    uint32_t pay_done =
-        s->tx.done >= sizeof(ChunkHdr) ? s->tx.done - 16 : 0;
+    // Build the iovec for the unwritten tail of the batch's virtual stream
+    // [hdr0|pay0|hdr1|pay1|...].
     int iovn = 0;
-    if (hdr_left)
-      s->ur.iov[iovn++] = {(char*)&s->tx.hdr + (16 - hdr_left), hdr_left};
-    if (s->tx.hdr.len - pay_done)
-      s->ur.iov[iovn++] = {(void*)(s->tx.payload + pay_done),
-                           s->tx.hdr.len - pay_done};
-    if (iovn == 0) {  // nothing left (shouldn't happen)
-      complete_send_chunk(s);
+    uint32_t skip = s->ur.done;
+    for (int i = 0; i < s->ur.nchunks; i++) {
+      uint32_t span = 16 + s->ur.hdrs[i].len;
+      if (skip >= span) {
+        skip -= span;
+        continue;
+      }
+      uint32_t o = skip;
+      skip = 0;
+      if (o < 16)
+        s->ur.iov[iovn++] = {(char*)&s->ur.hdrs[i] + o, 16 - o};
+      uint32_t pay_off = o > 16 ? o - 16 : 0;
+      if (s->ur.hdrs[i].len > pay_off)
+        s->ur.iov[iovn++] = {(void*)(s->ur.payloads[i] + pay_off),
+                             s->ur.hdrs[i].len - pay_off};
+    }
+    if (iovn == 0) {
+      finish_batch(s);
       return;
     }
     io_uring_sqe* sqe = ring_.get_sqe();
@@ -251,18 +270,21 @@ class UringIoThread : public IIoThread {
     s->ur.op = 1;
   }
 
-  void complete_send_chunk(TcpSock* s) {
+  void finish_batch(TcpSock* s) {
     SendComm* c = s->scomm;
-    SendRequest* r = s->tx.req;
-    uint32_t len = s->tx.hdr.len;
-    s->tx.active = false;
-    s->tx.req = nullptr;
-    if (r->total == 0) {
-      r->hdr_sent.store(true, std::memory_order_release);
-    } else {
-      c->stats.bytes_sent.fetch_add(len, std::memory_order_relaxed);
-      r->sent.fetch_add(len, std::memory_order_acq_rel);
+    for (int i = 0; i < s->ur.nchunks; i++) {
+      SendRequest* r = s->ur.reqs[i];
+      uint32_t len = s->ur.hdrs[i].len;
+      if (r->total == 0) {
+        r->hdr_sent.store(true, std::memory_order_release);
+      } else {
+        c->stats.bytes_sent.fetch_add(len, std::memory_order_relaxed);
+        r->sent.fetch_add(len, std::memory_order_acq_rel);
+      }
     }
+    s->ur.nchunks = 0;
+    s->ur.done = 0;
+    s->ur.batch_bytes = 0;
   }
 
   void on_send_cqe(TcpSock* s, int res) {
@@ -277,9 +299,9 @@ class UringIoThread : public IIoThread {
       BNET_WARN("bnet(uring) send error: %s", strerror(-res));
       return;
     }
-    s->tx.done += (uint32_t)res;
-    if (s->tx.done == 16 + s->tx.hdr.len) complete_send_chunk(s);
-    submit_send(s);  // next span / next chunk
+    s->ur.done += (uint32_t)res;
+    if (s->ur.done == s->ur.batch_bytes) finish_batch(s);
+    submit_send(s);  // remainder / next batch
   }
 
   // ---- recv path -------------------------------------------------------
@@ -416,10 +438,14 @@ class UringIoThread : public IIoThread {
         }
         case Task::KICK:
           if (std::find(socks_.begin(), socks_.end(), t.s) != socks_.end()) {
-            if (t.s->is_recv)
-              resume_parked(t.s);
-            else
+            if (t.s->is_recv) {
+              if (t.s->parked.load(std::memory_order_relaxed))
+                resume_parked(t.s);
+              else
+                submit_recv(t.s);  // no-op if an op is already in flight
+            } else {
               submit_send(t.s);
+            }
           }
           break;
       }
@@ -477,7 +503,8 @@ class UringIoThread : public IIoThread {
       if (spin || had) {
         // jobs/watermarks may have advanced: retry idle senders
         for (TcpSock* s : socks_)
-          if (!s->is_recv && !s->ur.op && !s->tx.active) submit_send(s);
+          if (!s->is_recv && !s->ur.op && s->ur.nchunks == 0)
+            submit_send(s);
         ring_.enter(0);
       }
     }
