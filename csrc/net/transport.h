@@ -161,7 +161,6 @@ struct TcpSock {
     uint32_t batch_bytes = 0;  // sum of (16 + len) over the batch
     uint32_t done = 0;         // bytes of the batch written so far
     struct iovec iov[8];
-    uint32_t want = 0;  // bytes requested by the outstanding recv
   } ur;
   std::atomic<bool> parked{false};  // recv: waiting for a not-yet-posted seq
   SendComm* scomm = nullptr;

@@ -329,7 +329,6 @@ class UringIoThread : public IIoThread {
     sqe->user_data = (uint64_t)(uintptr_t)s;
     ring_.advance_tail();
     s->ur.op = 2;
-    s->ur.want = want;
   }
 
   void on_recv_cqe(TcpSock* s, int res) {

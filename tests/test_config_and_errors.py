@@ -145,3 +145,66 @@ def test_peer_death_surfaces_error():
     result = qr.get(timeout=90)
     pr.join(30)
     assert result.startswith("error-surfaced"), result
+
+
+def _garbage_then_legit(env, q):
+    for k, v in env.items():
+        os.environ[k] = v
+    import ctypes as C
+    import socket
+    import struct
+    import sys
+    import time
+
+    sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+    from baguanet.plugin import Plugin
+
+    p = Plugin()
+    handle, lcomm = p.listen(0)
+    # decode the port from the handle: {u32 magic, u16 family, u16 port_be}
+    magic, family, port_be = struct.unpack_from("<IHH", bytes(handle))
+    port = socket.ntohs(port_be)
+    # throw garbage at the listener: wrong magic, short writes, instant close
+    for payload in (b"GET / HTTP/1.0\r\n\r\n", b"\x00" * 3, b""):
+        g = socket.create_connection(("127.0.0.1", port), timeout=5)
+        if payload:
+            g.sendall(payload)
+        g.close()
+        # pump accept so it processes the junk
+        t0 = time.monotonic()
+        while time.monotonic() - t0 < 0.2:
+            assert p.accept(lcomm) is None
+    # a legitimate connection must still succeed
+    scomm = rcomm = None
+    t0 = time.monotonic()
+    while scomm is None or rcomm is None:
+        assert time.monotonic() - t0 < 30
+        if scomm is None:
+            scomm = p.connect(0, handle)
+        if rcomm is None:
+            rcomm = p.accept(lcomm)
+    buf = C.create_string_buffer(b"hello-after-garbage", 19)
+    rbuf = C.create_string_buffer(20)
+    mh = p.reg_mr(scomm, None, 0)
+    rreq = p.irecv(rcomm, rbuf, 19, mh)
+    sreq = p.isend(scomm, buf, 19, mh)
+    assert p.wait(sreq, 30) == 19 and p.wait(rreq, 30) == 19
+    assert rbuf.raw[:19] == b"hello-after-garbage"
+    p.close_send(scomm)
+    p.close_recv(rcomm)
+    p.close_listen(lcomm)
+    q.put("ok")
+
+
+def test_garbage_connections_rejected():
+    """Junk connections to a listener must not break legitimate accepts."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    proc = ctx.Process(
+        target=_garbage_then_legit,
+        args=({"NCCL_SOCKET_IFNAME": "lo"}, q),
+    )
+    proc.start()
+    assert q.get(timeout=120) == "ok"
+    proc.join(30)
+    assert proc.exitcode == 0
