@@ -40,7 +40,10 @@ def main():
     smh_h = p.reg_mr(scomm, None, 0)
     rmh_h = p.reg_mr(rcomm, None, 0)
 
-    inflight = []  # (sreq, rreq, verify_fn)
+    # entries: dict(sreq, rreq, sdone, rdone, verify, nbytes) — a request is
+    # tested ONLY until it reports done (test() frees the slot at done;
+    # re-testing a freed handle is outside the ABI contract)
+    inflight = []
     sent = 0
     bytes_total = 0
     t0 = time.monotonic()
@@ -74,7 +77,9 @@ def main():
                     torch.cuda.synchronize()
                     assert torch.equal(src, dst), "GPU payload corrupt"
 
-                inflight.append((sreq, rreq, verify, nbytes))
+                inflight.append(dict(sreq=sreq, rreq=rreq, sdone=False,
+                                     rdone=False, verify=verify,
+                                     nbytes=nbytes))
             else:
                 payload = rng.randbytes(size) if size else b""
                 sbuf = C.create_string_buffer(payload, max(size, 1))
@@ -86,22 +91,21 @@ def main():
                 def verify(payload=payload, rbuf=rbuf, sbuf=sbuf, size=size):
                     assert rbuf.raw[:size] == payload, "payload corrupt"
 
-                inflight.append((sreq, rreq, verify, size))
+                inflight.append(dict(sreq=sreq, rreq=rreq, sdone=False,
+                                     rdone=False, verify=verify,
+                                     nbytes=size))
             sent += 1
         done_any = False
         for item in list(inflight):
-            sreq, rreq, verify, nbytes = item
-            sd, _ = p.test(sreq) if sreq is not None else (True, 0)
-            rd, _ = p.test(rreq)
-            if sd:
-                item_idx = inflight.index(item)
-                if rd:
-                    verify()
-                    bytes_total += nbytes
-                    inflight.pop(item_idx)
-                    done_any = True
-                else:
-                    inflight[item_idx] = (None, rreq, verify, nbytes)
+            if not item["sdone"]:
+                item["sdone"], _ = p.test(item["sreq"])
+            if not item["rdone"]:
+                item["rdone"], _ = p.test(item["rreq"])
+            if item["sdone"] and item["rdone"]:
+                item["verify"]()
+                bytes_total += item["nbytes"]
+                inflight.remove(item)
+                done_any = True
         if not done_any:
             time.sleep(0)
     dt = time.monotonic() - t0
