@@ -380,14 +380,22 @@ void IoThread::progress_recv(TcpSock* s) {
       int hr = process_recv_header(s);
       if (hr < 0) return;  // protocol error (comm error set)
       if (hr == 1) {
-        // not posted yet — park; irecv() will kick us
-        if (!s->parked.exchange(true)) {
+        // Not posted yet — park.  Dekker-style handshake with irecv():
+        // we publish `parked` (seq_cst) BEFORE re-checking the slot;
+        // irecv publishes the slot (seq_cst) BEFORE checking `parked`.
+        // At least one side must see the other, so either we proceed now
+        // or the kick finds parked==true.
+        s->parked.store(true, std::memory_order_seq_cst);
+        hr = process_recv_header(s);
+        if (hr < 0) return;
+        if (hr == 1) {
           epoll_event ev{};
           ev.events = 0;
           ev.data.ptr = s;
           epoll_ctl(epfd_, EPOLL_CTL_MOD, s->fd, &ev);
+          return;
         }
-        return;
+        // posted concurrently: fall through and continue
       }
       if (s->parked.exchange(false)) {
         epoll_event ev{};
@@ -530,12 +538,13 @@ void Engine::kick_sock(TcpSock* s) {
 }
 
 void Engine::kick_comm(RecvComm* c) {
-  // UNCONDITIONAL: testing s->parked here races with the IO thread that
-  // is just deciding to park (it checked the slot before this irecv was
-  // published, we check parked before it sets the flag — both miss, and
-  // the socket sleeps forever with EPOLLIN disarmed).  A kick to an
-  // unparked socket is a cheap no-op read().
-  for (TcpSock* s : c->socks) threads_[s->io_thread]->kick(s);
+  // Only parked sockets need a wakeup; the park/post race is closed by
+  // the seq_cst publish-then-recheck handshake (see progress_recv): the
+  // parking thread re-checks the slot after publishing `parked`, and
+  // irecv publishes the slot before loading `parked` here.
+  for (TcpSock* s : c->socks)
+    if (s->parked.load(std::memory_order_seq_cst))
+      threads_[s->io_thread]->kick(s);
 }
 
 }  // namespace baguanet

@@ -363,8 +363,12 @@ class UringIoThread : public IIoThread {
         int hr = process_recv_header(s);
         if (hr < 0) return;
         if (hr == 1) {
-          s->parked.store(true);  // irecv() kick resumes us
-          return;
+          // same publish-then-recheck handshake as the epoll engine
+          s->parked.store(true, std::memory_order_seq_cst);
+          hr = process_recv_header(s);
+          if (hr < 0) return;
+          if (hr == 1) return;  // truly parked; irecv() kick resumes us
+          s->parked.store(false, std::memory_order_relaxed);
         }
         if (s->rx.hdr.len == 0) finish_rx_chunk(s);
       }
