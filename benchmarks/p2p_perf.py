@@ -28,7 +28,12 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 DEPTH = 16  # outstanding messages (NCCL proxy pipelines similarly)
 
 
-def _set_env(args):
+def _set_env(args, role=None):
+    # per-role engine override for A/B isolation (BNET_IMPL_SENDER/RECEIVER)
+    if role:
+        impl = os.environ.get(f"BNET_IMPL_{role.upper()}")
+        if impl:
+            os.environ["BNET_IMPLEMENT"] = impl
     os.environ["NCCL_SOCKET_IFNAME"] = args.ifname
     os.environ["BNET_NSTREAMS"] = str(args.nstreams)
     os.environ["BNET_MIN_CHUNKSIZE"] = str(args.min_chunk)
@@ -36,7 +41,7 @@ def _set_env(args):
 
 
 def _receiver(conn, args, out_q):
-    _set_env(args)
+    _set_env(args, "receiver")
     from baguanet.plugin import Plugin
 
     p = Plugin()
@@ -73,7 +78,7 @@ def _receiver(conn, args, out_q):
 
 
 def _sender(conn, args, out_q):
-    _set_env(args)
+    _set_env(args, "sender")
     from baguanet.plugin import Plugin
 
     p = Plugin()

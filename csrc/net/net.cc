@@ -428,9 +428,11 @@ ncclResult_t Net::irecv(void* recv_comm, int n, void** data, int* sizes,
     r->stage = nullptr;
   }
   uint32_t seq = c->post_next;
-  // seq_cst: pairs with the parking IO thread's publish-then-recheck
-  // (transport.cc progress_recv) so a concurrent park cannot be missed
+  // seq_cst store + seq_cst fence before kick_comm's parked load: pairs
+  // with the parking IO thread's publish-then-recheck (drain_recv) so a
+  // concurrent park cannot be missed
   r->state_seq.store(pack_ss(seq, REQ_ACTIVE), std::memory_order_seq_cst);
+  std::atomic_thread_fence(std::memory_order_seq_cst);
   c->post_next++;
   c->stats.irecv_count.fetch_add(1, std::memory_order_relaxed);
   auto& T = Telemetry::get();

@@ -341,9 +341,14 @@ void IoThread::progress_send(TcpSock* s) {
   }
 }
 
-void IoThread::progress_recv(TcpSock* s) {
+// Greedy nonblocking recv drain shared by both engines: loop the rx state
+// machine with plain read() until the kernel buffer is empty.  Returns:
+//   RX_WAIT   — EAGAIN: caller waits for readability (epoll arm / RECV op)
+//   RX_PARKED — a chunk for an un-posted request arrived; irecv() kicks
+//   RX_CLOSED — error (comm error set) or benign end-of-stream EOF
+int drain_recv(TcpSock* s) {
   RecvComm* c = s->rcomm;
-  if (c->error.load(std::memory_order_relaxed)) return;
+  if (c->error.load(std::memory_order_relaxed)) return RX_CLOSED;
   while (true) {
     if (!s->rx.in_payload) {
       // read header
@@ -354,7 +359,8 @@ void IoThread::progress_recv(TcpSock* s) {
           s->rx.hdr_got += (uint32_t)n;
           continue;
         }
-        if (n < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) return;
+        if (n < 0 && (errno == EAGAIN || errno == EWOULDBLOCK))
+          return RX_WAIT;
         if (n < 0 && errno == EINTR) continue;
         if (n == 0 && s->rx.hdr_got == 0) {
           // EOF on a message boundary: benign iff no recv is pending
@@ -363,22 +369,16 @@ void IoThread::progress_recv(TcpSock* s) {
             if (ss_state(r.state_seq.load(std::memory_order_acquire)) ==
                 REQ_ACTIVE)
               pending = true;
-          if (!pending) {
-            epoll_event ev{};
-            ev.events = 0;
-            ev.data.ptr = s;
-            epoll_ctl(epfd_, EPOLL_CTL_MOD, s->fd, &ev);
-            return;
-          }
+          if (!pending) return RX_CLOSED;
         }
         c->error.store(n == 0 ? ECONNRESET : (errno ? errno : EIO));
         BNET_WARN("bnet recv socket %s", n == 0 ? "eof mid-protocol"
                                                 : strerror(errno));
-        return;
+        return RX_CLOSED;
       }
       // header complete → locate the posted request
       int hr = process_recv_header(s);
-      if (hr < 0) return;  // protocol error (comm error set)
+      if (hr < 0) return RX_CLOSED;  // protocol error (comm error set)
       if (hr == 1) {
         // Not posted yet — park.  Dekker-style handshake with irecv():
         // we publish `parked` (seq_cst) BEFORE re-checking the slot;
@@ -386,23 +386,17 @@ void IoThread::progress_recv(TcpSock* s) {
         // At least one side must see the other, so either we proceed now
         // or the kick finds parked==true.
         s->parked.store(true, std::memory_order_seq_cst);
+        // the re-check load below is acquire; without a seq_cst fence it
+        // may be ordered before the store above (store-buffer litmus) and
+        // both sides can miss each other
+        std::atomic_thread_fence(std::memory_order_seq_cst);
         hr = process_recv_header(s);
-        if (hr < 0) return;
-        if (hr == 1) {
-          epoll_event ev{};
-          ev.events = 0;
-          ev.data.ptr = s;
-          epoll_ctl(epfd_, EPOLL_CTL_MOD, s->fd, &ev);
-          return;
-        }
+        if (hr < 0) return RX_CLOSED;
+        if (hr == 1) return RX_PARKED;
         // posted concurrently: fall through and continue
       }
-      if (s->parked.exchange(false)) {
-        epoll_event ev{};
-        ev.events = EPOLLIN;
-        ev.data.ptr = s;
-        epoll_ctl(epfd_, EPOLL_CTL_MOD, s->fd, &ev);
-      }
+      if (s->parked.load(std::memory_order_relaxed))
+        s->parked.store(false, std::memory_order_relaxed);
       if (s->rx.hdr.len == 0) {
         // empty chunk (only for zero-byte messages)
         finish_rx_chunk(s);
@@ -416,14 +410,37 @@ void IoThread::progress_recv(TcpSock* s) {
         s->rx.remaining -= (uint32_t)n;
         continue;
       }
-      if (n < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) return;
+      if (n < 0 && (errno == EAGAIN || errno == EWOULDBLOCK))
+        return RX_WAIT;
       if (n < 0 && errno == EINTR) continue;
       c->error.store(n == 0 ? ECONNRESET : (errno ? errno : EIO));
       BNET_WARN("bnet recv payload %s", n == 0 ? "eof mid-chunk"
                                                : strerror(errno));
-      return;
+      return RX_CLOSED;
     }
     finish_rx_chunk(s);
+  }
+}
+
+void IoThread::progress_recv(TcpSock* s) {
+  bool was_parked = s->parked.load(std::memory_order_relaxed);
+  int rc = drain_recv(s);
+  if (rc == RX_WAIT) {
+    if (was_parked && !s->parked.load(std::memory_order_relaxed)) {
+      epoll_event ev{};  // resumed from parked: re-arm EPOLLIN
+      ev.events = EPOLLIN;
+      ev.data.ptr = s;
+      epoll_ctl(epfd_, EPOLL_CTL_MOD, s->fd, &ev);
+    }
+    return;
+  }
+  if (rc == RX_PARKED || rc == RX_CLOSED) {
+    // stop polling this socket (parked: irecv's kick resumes; closed:
+    // nothing more to read — errors surface via the comm error)
+    epoll_event ev{};
+    ev.events = 0;
+    ev.data.ptr = s;
+    epoll_ctl(epfd_, EPOLL_CTL_MOD, s->fd, &ev);
   }
 }
 

@@ -256,6 +256,9 @@ struct ListenComm {
 // Shared transport state-machine helpers (used by both engines).
 SendRequest* claim_chunk(SendComm* c, int sock_idx, uint32_t* off,
                          uint32_t* len);
+enum RxResult { RX_WAIT = 0, RX_PARKED = 1, RX_CLOSED = 2 };
+// Greedy nonblocking drain of the rx state machine (see transport.cc).
+int drain_recv(TcpSock* s);
 // rx.hdr is complete: locate/validate the posted request and prime
 // rx.target/remaining.  Returns 0 = proceed, 1 = park (request not yet
 // posted), -1 = protocol error (comm error set).

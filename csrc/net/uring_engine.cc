@@ -357,38 +357,24 @@ class UringIoThread : public IIoThread {
       BNET_WARN("bnet(uring) recv eof mid-protocol");
       return;
     }
+    // absorb the CQE's bytes into the rx state, then drain the socket
+    // greedily with plain read() — one RECV op per burst of buffered
+    // data (a RECV op per span left the kernel buffer idle between
+    // completion and resubmission: 2-3x slower at 1 MiB, measured)
     if (!s->rx.in_payload) {
       s->rx.hdr_got += (uint32_t)res;
-      if (s->rx.hdr_got == sizeof(ChunkHdr)) {
-        int hr = process_recv_header(s);
-        if (hr < 0) return;
-        if (hr == 1) {
-          // same publish-then-recheck handshake as the epoll engine
-          s->parked.store(true, std::memory_order_seq_cst);
-          hr = process_recv_header(s);
-          if (hr < 0) return;
-          if (hr == 1) return;  // truly parked; irecv() kick resumes us
-          s->parked.store(false, std::memory_order_relaxed);
-        }
-        if (s->rx.hdr.len == 0) finish_rx_chunk(s);
-      }
     } else {
       s->rx.target += res;
       s->rx.remaining -= (uint32_t)res;
       if (s->rx.remaining == 0) finish_rx_chunk(s);
     }
-    submit_recv(s);
+    if (drain_recv(s) == RX_WAIT) submit_recv(s);
+    // RX_PARKED: irecv()'s kick resumes; RX_CLOSED: stop re-arming
   }
 
   void resume_parked(TcpSock* s) {
-    if (!s->parked.load(std::memory_order_relaxed)) return;
-    // header already buffered in rx.hdr — retry matching
-    int hr = process_recv_header(s);
-    if (hr < 0) return;
-    if (hr == 1) return;  // still unposted
-    s->parked.store(false);
-    if (s->rx.hdr.len == 0) finish_rx_chunk(s);
-    submit_recv(s);
+    // header already buffered in rx.hdr — drain retries the match
+    if (drain_recv(s) == RX_WAIT) submit_recv(s);
   }
 
   void handle_tasks() {
