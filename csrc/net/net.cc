@@ -370,7 +370,10 @@ ncclResult_t Net::isend(void* send_comm, void* data, int size, int tag,
     r->avail.store((uint32_t)size, std::memory_order_relaxed);
   }
   uint32_t seq = c->seq_next;
-  r->state_seq.store(pack_ss(seq, REQ_ACTIVE), std::memory_order_release);
+  // seq_cst store + fence pairs with the sender sockets' idle
+  // publish-then-recheck (progress_send) — see kick_comm
+  r->state_seq.store(pack_ss(seq, REQ_ACTIVE), std::memory_order_seq_cst);
+  std::atomic_thread_fence(std::memory_order_seq_cst);
   c->seq_next++;
   c->stats.isend_count.fetch_add(1, std::memory_order_relaxed);
   auto& T = Telemetry::get();

@@ -282,8 +282,16 @@ void IoThread::progress_send(TcpSock* s) {
       uint32_t off = 0, len = 0;
       SendRequest* r = claim_chunk(c, s->idx, &off, &len);
       if (!r) {
-        set_epollout(s, false);
-        return;
+        // going idle: publish-then-recheck so a concurrent isend either
+        // sees snd_idle (and kicks) or we see its request now
+        s->snd_idle.store(true, std::memory_order_seq_cst);
+        std::atomic_thread_fence(std::memory_order_seq_cst);
+        r = claim_chunk(c, s->idx, &off, &len);
+        if (!r) {
+          set_epollout(s, false);
+          return;
+        }
+        s->snd_idle.store(false, std::memory_order_relaxed);
       }
       s->tx.active = true;
       s->tx.req = r;
@@ -537,21 +545,24 @@ void Engine::unregister_sock_sync(TcpSock* s) {
 }
 
 void Engine::kick_comm(SendComm* c, int max_socks) {
-  // Kicking more sockets than the message has chunks just burns eventfd
-  // wakeups (hurts small-message rates).  Kicks are UNCONDITIONAL for the
-  // chosen sockets: skipping an apparently-busy socket races with its
-  // transition to idle (it may have scanned for chunks before the new
-  // request was published and never look again) — a missed wakeup hangs
-  // the message.
+  // Kick only IDLE sockets (snd_idle), up to the message's chunk count —
+  // a busy socket keeps claiming work itself, and its idle transition is
+  // covered by the publish-then-recheck handshake in progress_send /
+  // submit_send (it re-scans after setting snd_idle, so either it finds
+  // this request or this kick sees snd_idle).
   int n = max_socks < 0 ? (int)c->socks.size() : max_socks;
   for (TcpSock* s : c->socks) {
-    if (n-- <= 0) break;
-    threads_[s->io_thread]->kick(s);
+    if (n <= 0) break;
+    if (s->snd_idle.load(std::memory_order_seq_cst)) {
+      threads_[s->io_thread]->kick(s);
+      n--;
+    }
   }
 }
 
 void Engine::kick_sock(TcpSock* s) {
-  threads_[s->io_thread]->kick(s);  // unconditional — see kick_comm
+  if (s->snd_idle.load(std::memory_order_seq_cst))
+    threads_[s->io_thread]->kick(s);  // handshake-guarded — see kick_comm
 }
 
 void Engine::kick_comm(RecvComm* c) {

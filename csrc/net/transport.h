@@ -148,6 +148,11 @@ struct TcpSock {
   int idx = 0;  // position within its comm's socket list
   bool is_recv = false;
   bool want_epollout = false;
+  // send-side idle flag for the kick fast path: set (seq_cst) by the IO
+  // thread right before its final failed claim re-check; isend publishes
+  // the request (seq_cst + fence) then kicks only sockets with snd_idle
+  // set — same Dekker publish-then-recheck handshake as recv parking
+  std::atomic<bool> snd_idle{false};
   // io_uring engine per-socket state (unused by the epoll engine)
   static constexpr int kUrBatch = 4;  // chunks per WRITEV submission
   struct {

@@ -226,6 +226,13 @@ class UringIoThread : public IIoThread {
       while (s->ur.nchunks < TcpSock::kUrBatch) {
         uint32_t off = 0, len = 0;
         SendRequest* r = claim_chunk(c, s->idx, &off, &len);
+        if (!r && s->ur.nchunks == 0) {
+          // same idle publish-then-recheck handshake as the epoll engine
+          s->snd_idle.store(true, std::memory_order_seq_cst);
+          std::atomic_thread_fence(std::memory_order_seq_cst);
+          r = claim_chunk(c, s->idx, &off, &len);
+          if (r) s->snd_idle.store(false, std::memory_order_relaxed);
+        }
         if (!r) break;
         int i = s->ur.nchunks++;
         s->ur.reqs[i] = r;
@@ -234,7 +241,8 @@ class UringIoThread : public IIoThread {
         s->ur.payloads[i] = r->src + off;
         s->ur.batch_bytes += 16 + len;
       }
-      if (s->ur.nchunks == 0) return;  // idle until next kick
+      if (s->ur.nchunks == 0) return;  // idle (snd_idle set) until kicked
+    s->snd_idle.store(false, std::memory_order_relaxed);
     }
     // Build the iovec for the unwritten tail of the batch's virtual stream
     // [hdr0|pay0|hdr1|pay1|...].
