@@ -3,6 +3,7 @@
 #include "staging.h"
 
 #include <hip/hip_runtime.h>
+#include <rocprofiler-sdk-roctx/roctx.h>
 
 #include <algorithm>
 #include <mutex>
@@ -154,12 +155,24 @@ static void issue_copy(void* dst, const void* src, size_t n, hipMemcpyKind k,
   }
 }
 
+// roctx range markers make the staging pipeline visible in rocprofv3
+// traces ("add rocprof markers around the HIP staging path").
+static bool roctx_on() {
+  static int on = [] {
+    const char* v = getenv("BNET_ROCTX");
+    return v && *v == '1' ? 1 : 0;
+  }();
+  return on == 1;
+}
+
 bool stage_send_begin(StagePool* p, SendRequest* req, const void* src,
                       uint32_t total) {
+  if (roctx_on()) roctxRangePush("bnet_stage_send_d2h");
   std::lock_guard<std::mutex> lk(p->mu);
   auto* a = new StageAlloc();
   if (!(a->host = p->alloc(std::max(total, 1u), &a->pool_off))) {
     delete a;
+    if (roctx_on()) roctxRangePop();
     return false;
   }
   a->size = std::max(total, 1u);
@@ -182,6 +195,7 @@ bool stage_send_begin(StagePool* p, SendRequest* req, const void* src,
     p->inflight.push_back(a);
     p->pending.fetch_add(1, std::memory_order_release);
   }
+  if (roctx_on()) roctxRangePop();
   return true;
 }
 
@@ -239,6 +253,7 @@ void stage_recv_issue(RecvRequest* req, uint32_t offset, uint32_t len) {
   StageAlloc* a = (StageAlloc*)req->stage;
   StagePool* p = a->pool;
   if (!len) return;
+  if (roctx_on()) roctxMarkA("bnet_stage_recv_h2d");
   std::lock_guard<std::mutex> lk(p->mu);
   issue_copy(a->gpu_dst + offset, a->host + offset, len,
              hipMemcpyHostToDevice, p->h2d);
