@@ -91,13 +91,6 @@ void IoThread::kick(TcpSock* s) {
   (void)!write(evfd_, &one, sizeof(one));
 }
 
-void IoThread::kick_all() {
-  std::lock_guard<std::mutex> lk(task_mu_);
-  tasks_.push_back({Task::KICKALL, nullptr, nullptr, nullptr});
-  uint64_t one = 1;
-  (void)!write(evfd_, &one, sizeof(one));
-}
-
 void IoThread::handle_tasks() {
   std::vector<Task> batch;
   {
@@ -136,9 +129,6 @@ void IoThread::handle_tasks() {
         // The socket may have been removed between enqueue and drain.
         if (std::find(socks_.begin(), socks_.end(), t.s) != socks_.end())
           progress(t.s);
-        break;
-      case Task::KICKALL:
-        for (TcpSock* s : socks_) progress(s);
         break;
     }
   }

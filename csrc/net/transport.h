@@ -288,7 +288,6 @@ class IoThread : public IIoThread {
   void add_sock(TcpSock* s) override;
   void remove_sock_sync(TcpSock* s) override;
   void kick(TcpSock* s) override;
-  void kick_all();
 
  private:
   void run();
@@ -304,7 +303,7 @@ class IoThread : public IIoThread {
   std::atomic<bool> stop_{false};
   std::mutex task_mu_;
   struct Task {
-    enum { ADD, REMOVE, KICK, KICKALL } kind;
+    enum { ADD, REMOVE, KICK } kind;
     TcpSock* s;
     std::condition_variable* cv;
     bool* flag;
