@@ -367,6 +367,7 @@ ncclResult_t Net::isend(void* send_comm, void* data, int size, int tag,
     r->avail.store(0, std::memory_order_relaxed);
   } else {
     r->src = (const char*)data;
+    r->stage = nullptr;
     r->avail.store((uint32_t)size, std::memory_order_relaxed);
   }
   uint32_t seq = c->seq_next;
