@@ -154,18 +154,18 @@ struct TcpSock {
   // set — same Dekker publish-then-recheck handshake as recv parking
   std::atomic<bool> snd_idle{false};
   // io_uring engine per-socket state (unused by the epoll engine)
-  static constexpr int kUrBatch = 4;  // chunks per WRITEV submission
+  static constexpr int kUrBatch = 8;  // chunks per WRITEV submission
   struct {
     uint8_t op = 0;  // 0 none, 1 send(writev), 2 recv
     bool closing = false;  // removal in progress: no resubmission
     // batched send: up to kUrBatch chunks in one ordered WRITEV
     int nchunks = 0;
-    ChunkHdr hdrs[4];
-    const char* payloads[4];
-    SendRequest* reqs[4];
+    ChunkHdr hdrs[8];
+    const char* payloads[8];
+    SendRequest* reqs[8];
     uint32_t batch_bytes = 0;  // sum of (16 + len) over the batch
     uint32_t done = 0;         // bytes of the batch written so far
-    struct iovec iov[8];
+    struct iovec iov[16];
   } ur;
   std::atomic<bool> parked{false};  // recv: waiting for a not-yet-posted seq
   SendComm* scomm = nullptr;
