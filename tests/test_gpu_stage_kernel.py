@@ -58,3 +58,43 @@ def test_kernel_staging_roundtrip():
     assert q.get(timeout=240) == "ok"
     proc.join(30)
     assert proc.exitcode == 0
+
+
+def _oversize(q):
+    os.environ["NCCL_SOCKET_IFNAME"] = "lo"
+    os.environ["BNET_STAGE_POOL"] = str(8 << 20)  # 8 MiB pool
+    import ctypes as C
+    import sys
+
+    sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+    sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    import torch
+
+    from baguanet.plugin import Plugin
+    from test_plugin_loopback import establish
+
+    p = Plugin()
+    lcomm, scomm, rcomm = establish(p)
+    big = torch.zeros(16 << 18, device="cuda")  # 16 MiB > pool
+    torch.cuda.synchronize()
+    mh = p.reg_mr(scomm, C.c_void_p(big.data_ptr()), big.numel() * 4, 0x2)
+    try:
+        p.isend(scomm, C.c_void_p(big.data_ptr()), big.numel() * 4, mh)
+        q.put("no-error")
+    except RuntimeError:
+        q.put("errored")  # ncclInternalError, not an infinite retry
+    p.close_send(scomm)
+    p.close_recv(rcomm)
+    p.close_listen(lcomm)
+
+
+def test_message_larger_than_pool_errors():
+    """A single message larger than the staging pool must fail loudly
+    (ncclInternalError) instead of retrying forever."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    proc = ctx.Process(target=_oversize, args=(q,))
+    proc.start()
+    assert q.get(timeout=240) == "errored"
+    proc.join(30)
+    assert proc.exitcode == 0
