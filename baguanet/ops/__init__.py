@@ -64,3 +64,9 @@ def multi_pack(flat: torch.Tensor, tensors: list[torch.Tensor]) -> None:
 
 def multi_unpack(flat: torch.Tensor, tensors: list[torch.Tensor]) -> None:
     _load().multi_unpack(flat, tensors)
+
+
+def fused_sgd(params, grads, momenta, lr, momentum=0.0, weight_decay=0.0,
+              nesterov=False) -> None:
+    _load().fused_sgd(params, grads, momenta, lr, momentum, weight_decay,
+                      nesterov)

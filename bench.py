@@ -78,6 +78,7 @@ def main():
         )
 
     from baguanet.models import resnet50, vgg16
+    from baguanet.optim import FusedSGD
     from baguanet.parallel import BucketedDDP
 
     torch.manual_seed(42 + rank)
@@ -86,7 +87,9 @@ def main():
     if args.channels_last:
         model = model.to(memory_format=torch.channels_last)
     model = BucketedDDP(model, bucket_cap_mb=50.0)
-    opt = torch.optim.SGD(model.module.parameters(), lr=0.01, momentum=0.9)
+    # fused multi-tensor SGD kernel on GPU (identical numerics to
+    # torch.optim.SGD — tests/test_gpu_fused_sgd.py); eager math on CPU
+    opt = FusedSGD(model.module.parameters(), lr=0.01, momentum=0.9)
 
     x = torch.randn(args.batch, 3, 224, 224, device=device)
     if args.channels_last:

@@ -16,11 +16,22 @@ struct PackDesc {
   uint32_t bytes;
 };
 
+struct SgdDesc {
+  float* p;
+  const float* g;
+  float* m;
+  uint32_t numel;
+};
+
 void launch_copy_kernel(void* dst, const void* src, size_t bytes,
                         hipStream_t stream);
 size_t multi_copy_launch(const PackDesc* host_descs, int ndesc, void* scratch,
                          size_t scratch_bytes, void* staging,
                          hipStream_t stream);
+size_t multi_sgd_launch(const SgdDesc* host_descs, int ndesc, void* scratch,
+                        size_t scratch_bytes, void* staging, float lr,
+                        float mu, float wd, int nesterov,
+                        hipStream_t stream);
 
 namespace {
 
@@ -104,6 +115,47 @@ void multi_unpack(torch::Tensor flat, std::vector<torch::Tensor> tensors) {
   run_multi(descs);
 }
 
+// Fused SGD step over all parameters in one launch (fp32, torch.optim.SGD
+// semantics).
+void fused_sgd(std::vector<torch::Tensor> params,
+               std::vector<torch::Tensor> grads,
+               std::vector<torch::Tensor> momenta, double lr, double momentum,
+               double weight_decay, bool nesterov) {
+  TORCH_CHECK(params.size() == grads.size() &&
+                  params.size() == momenta.size(),
+              "list size mismatch");
+  std::vector<SgdDesc> descs;
+  descs.reserve(params.size());
+  for (size_t i = 0; i < params.size(); i++) {
+    auto &p = params[i], &g = grads[i], &m = momenta[i];
+    TORCH_CHECK(p.is_cuda() && g.is_cuda() && m.is_cuda(),
+                "fused_sgd requires CUDA tensors");
+    TORCH_CHECK(p.scalar_type() == torch::kFloat32 &&
+                    g.scalar_type() == torch::kFloat32 &&
+                    m.scalar_type() == torch::kFloat32,
+                "fused_sgd is fp32-only");
+    TORCH_CHECK(p.is_contiguous() && g.is_contiguous() && m.is_contiguous(),
+                "contiguous tensors required");
+    TORCH_CHECK(p.numel() == g.numel() && p.numel() == m.numel(),
+                "numel mismatch");
+    descs.push_back({p.data_ptr<float>(), g.data_ptr<float>(),
+                     m.data_ptr<float>(), (uint32_t)p.numel()});
+  }
+  auto& s = scratch();
+  size_t need = multi_sgd_launch(descs.data(), (int)descs.size(), s.dev,
+                                 s.size, s.host, (float)lr, (float)momentum,
+                                 (float)weight_decay, nesterov ? 1 : 0,
+                                 current_stream());
+  if (need) {
+    s.ensure(need);
+    need = multi_sgd_launch(descs.data(), (int)descs.size(), s.dev, s.size,
+                            s.host, (float)lr, (float)momentum,
+                            (float)weight_decay, nesterov ? 1 : 0,
+                            current_stream());
+    TORCH_CHECK(need == 0, "multi_sgd_launch failed");
+  }
+}
+
 }  // namespace
 }  // namespace baguanet
 
@@ -114,4 +166,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused multi-tensor pack into flat buffer");
   m.def("multi_unpack", &baguanet::multi_unpack,
         "fused multi-tensor scatter from flat buffer");
+  m.def("fused_sgd", &baguanet::fused_sgd,
+        "fused multi-tensor SGD step (params, grads, momenta, lr, momentum, "
+        "weight_decay, nesterov)");
 }
