@@ -107,8 +107,6 @@ size_t multi_copy_launch(const PackDesc* host_descs, int ndesc, void* scratch,
   return 0;
 }
 
-}  // namespace baguanet
-
 // ---------------------------------------------------------------------------
 // Fused multi-tensor SGD (momentum, weight decay, nesterov) — the whole
 // model's parameter update in ONE launch.  Matches torch.optim.SGD
@@ -207,3 +205,5 @@ size_t multi_sgd_launch(const SgdDesc* host_descs, int ndesc, void* scratch,
                      nesterov);
   return 0;
 }
+
+}  // namespace baguanet
