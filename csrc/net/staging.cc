@@ -2,6 +2,8 @@
 
 #include "staging.h"
 
+#include "freelist.h"
+
 #include <hip/hip_runtime.h>
 #include <rocprofiler-sdk-roctx/roctx.h>
 
@@ -54,10 +56,7 @@ class StagePool {
   int device = 0;
   hipStream_t d2h = nullptr, h2d = nullptr;
   std::mutex mu;
-  struct Range {
-    size_t off, len;
-  };
-  std::vector<Range> free_list;
+  FreeList arena{0};
   std::vector<hipEvent_t> ev_cache;
   std::vector<StageAlloc*> inflight;  // send stagings with copies pending
   std::atomic<int> pending{0};
@@ -75,38 +74,12 @@ class StagePool {
   void put_event(hipEvent_t e) { ev_cache.push_back(e); }
 
   char* alloc(uint32_t sz, size_t* off_out) {
-    size_t need = (sz + 255) & ~size_t(255);
-    if (need == 0) need = 256;
-    for (auto it = free_list.begin(); it != free_list.end(); ++it) {
-      if (it->len >= need) {
-        size_t off = it->off;
-        it->off += need;
-        it->len -= need;
-        if (it->len == 0) free_list.erase(it);
-        *off_out = off;
-        return base + off;
-      }
-    }
-    return nullptr;
+    size_t off = arena.alloc(sz);
+    if (off == SIZE_MAX) return nullptr;
+    *off_out = off;
+    return base + off;
   }
-  void free(size_t off, uint32_t sz) {
-    size_t need = (sz + 255) & ~size_t(255);
-    if (need == 0) need = 256;
-    Range r{off, need};
-    auto it = std::lower_bound(
-        free_list.begin(), free_list.end(), r,
-        [](const Range& a, const Range& b) { return a.off < b.off; });
-    it = free_list.insert(it, r);
-    // coalesce with neighbors
-    if (it + 1 != free_list.end() && it->off + it->len == (it + 1)->off) {
-      it->len += (it + 1)->len;
-      free_list.erase(it + 1);
-    }
-    if (it != free_list.begin() && (it - 1)->off + (it - 1)->len == it->off) {
-      (it - 1)->len += it->len;
-      free_list.erase(it);
-    }
-  }
+  void free(size_t off, uint32_t sz) { arena.free(off, sz); }
 };
 
 bool staging_available() {
@@ -131,7 +104,7 @@ StagePool* stage_pool_create() {
   HIP_WARN(hipGetDevice(&p->device));
   HIP_WARN(hipStreamCreateWithFlags(&p->d2h, hipStreamNonBlocking));
   HIP_WARN(hipStreamCreateWithFlags(&p->h2d, hipStreamNonBlocking));
-  p->free_list.push_back({0, p->size});
+  p->arena = FreeList(p->size);
   return p;
 }
 
