@@ -18,6 +18,17 @@ sys.path.insert(0, os.path.join(os.path.dirname(os.path.dirname(
     os.path.abspath(__file__))), "tests"))
 
 
+def _dump_stall(inflight, rreq, sreq):
+    """A refused post with depth < 32 means some message stalled — dump
+    each in-flight entry's completion state before failing."""
+    print(f"POST REFUSED: rreq={rreq} sreq={sreq}; "
+          f"{len(inflight)} in flight:")
+    for i, it in enumerate(inflight):
+        print(f"  [{i}] nbytes={it['nbytes']} sdone={it['sdone']} "
+              f"rdone={it['rdone']}")
+    raise AssertionError("post refused — stalled message (state above)")
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--seconds", type=float, default=30.0)
@@ -71,7 +82,8 @@ def main():
                                rmh)
                 sreq = p.isend(scomm, C.c_void_p(src.data_ptr()), nbytes,
                                smh)
-                assert rreq is not None and sreq is not None
+                if rreq is None or sreq is None:
+                    _dump_stall(inflight, rreq, sreq)
 
                 def verify(src=src, dst=dst):
                     torch.cuda.synchronize()
@@ -86,7 +98,8 @@ def main():
                 rbuf = C.create_string_buffer(size + 1)
                 rreq = p.irecv(rcomm, rbuf, size, rmh_h)
                 sreq = p.isend(scomm, sbuf, size, smh_h)
-                assert rreq is not None and sreq is not None
+                if rreq is None or sreq is None:
+                    _dump_stall(inflight, rreq, sreq)
 
                 def verify(payload=payload, rbuf=rbuf, sbuf=sbuf, size=size):
                     assert rbuf.raw[:size] == payload, "payload corrupt"

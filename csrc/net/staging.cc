@@ -174,28 +174,38 @@ bool stage_send_begin(StagePool* p, SendRequest* req, const void* src,
 
 bool stage_poll(StagePool* p) {
   if (p->pending.load(std::memory_order_acquire) == 0) return false;
-  std::lock_guard<std::mutex> lk(p->mu);
+  SendComm* advanced_comm = nullptr;
   bool advanced = false;
-  for (auto it = p->inflight.begin(); it != p->inflight.end();) {
-    StageAlloc* a = *it;
-    while (a->events_done < a->events.size() &&
-           hipEventQuery(a->events[a->events_done]) == hipSuccess) {
-      a->events_done++;
-      advanced = true;
-      uint32_t avail =
-          std::min<uint64_t>((uint64_t)a->events_done * a->copy_chunk,
-                             a->total);
-      a->sreq->avail.store(avail, std::memory_order_release);
-    }
-    if (a->events_done == a->events.size()) {
-      for (auto e : a->events) p->put_event(e);
-      a->events.clear();
-      it = p->inflight.erase(it);
-      p->pending.fetch_sub(1, std::memory_order_release);
-    } else {
-      ++it;
+  {
+    std::lock_guard<std::mutex> lk(p->mu);
+    for (auto it = p->inflight.begin(); it != p->inflight.end();) {
+      StageAlloc* a = *it;
+      while (a->events_done < a->events.size() &&
+             hipEventQuery(a->events[a->events_done]) == hipSuccess) {
+        a->events_done++;
+        advanced = true;
+        uint32_t avail =
+            std::min<uint64_t>((uint64_t)a->events_done * a->copy_chunk,
+                               a->total);
+        a->sreq->avail.store(avail, std::memory_order_release);
+        advanced_comm = a->sreq->comm;
+      }
+      if (a->events_done == a->events.size()) {
+        for (auto e : a->events) p->put_event(e);
+        a->events.clear();
+        it = p->inflight.erase(it);
+        p->pending.fetch_sub(1, std::memory_order_release);
+      } else {
+        ++it;
+      }
     }
   }
+  // Watermark advances must WAKE idle claimer sockets: the advancing
+  // thread is not necessarily the one hosting the owning socket, and once
+  // `pending` hits zero other threads stop spin-retrying their senders —
+  // without this kick a just-completed staged send could sit unclaimed
+  // forever (observed as a rare soak stall).
+  if (advanced_comm) Engine::get().kick_comm(advanced_comm);
   return advanced;
 }
 
