@@ -129,8 +129,20 @@ class BucketedDDP(nn.Module):
         return self.module(*args, **kwargs)
 
     def finish_backward(self) -> None:
-        """Wait for all in-flight bucket all-reduces."""
+        """Wait for all in-flight bucket all-reduces.
+
+        Raises if a bucket only partially fired: some of its parameters
+        received no gradient this backward (unused parameters), which
+        would silently desynchronize ranks.
+        """
         for b in self.buckets:
+            if 0 < b.pending < len(b.params):
+                raise RuntimeError(
+                    f"BucketedDDP: {b.pending} of {len(b.params)} parameters"
+                    " in a bucket received no gradient this backward"
+                    " (unused parameters are not supported; ensure every"
+                    " parameter participates in the loss)"
+                )
             if b.work is not None:
                 b.work.wait()
                 b.work = None

@@ -148,3 +148,24 @@ def test_mixed_dtype_buckets():
         dts = {p.dtype for p in bkt.params}
         assert len(dts) == 1
         assert bkt.flat.dtype == next(iter(dts))
+
+
+def test_unused_parameter_raises():
+    import pytest
+
+    from baguanet.parallel import BucketedDDP
+
+    class M(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.used = torch.nn.Linear(4, 4)
+            self.unused = torch.nn.Linear(4, 4)
+
+        def forward(self, x):
+            return self.used(x)
+
+    m = BucketedDDP(M(), bucket_cap_mb=100, broadcast_params=False)
+    m.zero_grad()
+    m(torch.randn(2, 4)).sum().backward()
+    with pytest.raises(RuntimeError, match="received no gradient"):
+        m.finish_backward()
