@@ -445,6 +445,17 @@ void IoThread::progress_recv(TcpSock* s) {
 // rx.hdr is complete: locate/validate the posted request (see transport.h).
 int process_recv_header(TcpSock* s) {
   RecvComm* c = s->rcomm;
+  const ChunkHdr& h = s->rx.hdr;
+  // Frame sanity BEFORE slot matching: a corrupted header must error the
+  // comm, not park the socket waiting for a request that can never match.
+  constexpr uint32_t kMaxMsg = 1u << 30;  // NCCL's MAX_NET_SIZE
+  if (h.total > kMaxMsg || h.len > h.total || h.offset > h.total ||
+      h.offset + h.len > h.total || (h.len == 0 && h.total != 0)) {
+    BNET_WARN("bnet: malformed chunk header (seq=%u off=%u len=%u total=%u)",
+              h.seq, h.offset, h.len, h.total);
+    c->error.store(EPROTO);
+    return -1;
+  }
   uint32_t seq = s->rx.hdr.seq;
   RecvRequest* r = &c->reqs[seq % NCCL_NET_MAX_REQUESTS];
   if (r->state_seq.load(std::memory_order_acquire) !=

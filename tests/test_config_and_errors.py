@@ -208,3 +208,59 @@ def test_garbage_connections_rejected():
     assert q.get(timeout=120) == "ok"
     proc.join(30)
     assert proc.exitcode == 0
+
+
+def _malformed_frames(env, q):
+    for k, v in env.items():
+        os.environ[k] = v
+    import ctypes as C
+    import socket
+    import struct
+    import time
+
+    from baguanet.plugin import Plugin
+
+    p = Plugin()
+    handle, lcomm = p.listen(0)
+    magic, family, port_be = struct.unpack_from("<IHH", bytes(handle))
+    port = socket.ntohs(port_be)
+    # fake a single-stream sender: valid hello, then a malformed chunk
+    conn_id = 0xDEADBEEFCAFEF00D
+    hello = struct.pack("<IIQHHI", magic, 1, conn_id, 0, 1, 0)
+    g = socket.create_connection(("127.0.0.1", port), timeout=5)
+    g.sendall(hello)
+    rcomm = None
+    t0 = time.monotonic()
+    while rcomm is None and time.monotonic() - t0 < 30:
+        rcomm = p.accept(lcomm)
+    assert rcomm is not None
+    buf = C.create_string_buffer(4096)
+    mh = p.reg_mr(rcomm, buf, 4096)
+    req = p.irecv(rcomm, buf, 4096, mh)
+    assert req is not None
+    # header with len > total — must surface an error, not hang
+    g.sendall(struct.pack("<IIII", 0, 0, 4096, 16))
+    t0 = time.monotonic()
+    while time.monotonic() - t0 < 30:
+        try:
+            done, _ = p.test(req)
+        except RuntimeError:
+            q.put("errored")
+            g.close()
+            return
+        if done:
+            q.put("unexpected-done")
+            return
+        time.sleep(0.01)
+    q.put("timeout")
+
+
+def test_malformed_frame_errors_not_hangs():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    proc = ctx.Process(target=_malformed_frames,
+                       args=({"NCCL_SOCKET_IFNAME": "lo"}, q))
+    proc.start()
+    assert q.get(timeout=120) == "errored"
+    proc.join(30)
+    assert proc.exitcode == 0
