@@ -509,6 +509,9 @@ ncclResult_t Net::test(void* request, int* done, int* sizes) {
 
 ncclResult_t Net::close_send(void* send_comm) {
   auto* c = (SendComm*)send_comm;
+  BNET_INFO("close send comm %p: %llu isends, %llu bytes", send_comm,
+            (unsigned long long)c->stats.isend_count.load(),
+            (unsigned long long)c->stats.bytes_sent.load());
   for (auto* s : c->socks) {
     Engine::get().unregister_sock_sync(s);
     delete s;
@@ -520,6 +523,9 @@ ncclResult_t Net::close_send(void* send_comm) {
 
 ncclResult_t Net::close_recv(void* recv_comm) {
   auto* c = (RecvComm*)recv_comm;
+  BNET_INFO("close recv comm %p: %llu irecvs, %llu bytes", recv_comm,
+            (unsigned long long)c->stats.irecv_count.load(),
+            (unsigned long long)c->stats.bytes_recv.load());
   for (auto* s : c->socks) {
     Engine::get().unregister_sock_sync(s);
     delete s;
