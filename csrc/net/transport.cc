@@ -361,13 +361,9 @@ int drain_recv(TcpSock* s) {
           return RX_WAIT;
         if (n < 0 && errno == EINTR) continue;
         if (n == 0 && s->rx.hdr_got == 0) {
-          // EOF on a message boundary: benign iff no recv is pending
-          bool pending = false;
-          for (auto& r : c->reqs)
-            if (ss_state(r.state_seq.load(std::memory_order_acquire)) ==
-                REQ_ACTIVE)
-              pending = true;
-          if (!pending) return RX_CLOSED;
+          // EOF on a message boundary: benign iff no posted recv still
+          // expects socket bytes
+          if (!recv_socket_incomplete(c)) return RX_CLOSED;
         }
         c->error.store(n == 0 ? ECONNRESET : (errno ? errno : EIO));
         BNET_WARN("bnet recv socket %s", n == 0 ? "eof mid-protocol"
@@ -440,6 +436,16 @@ void IoThread::progress_recv(TcpSock* s) {
     ev.data.ptr = s;
     epoll_ctl(epfd_, EPOLL_CTL_MOD, s->fd, &ev);
   }
+}
+
+bool recv_socket_incomplete(RecvComm* c) {
+  for (auto& r : c->reqs) {
+    if (ss_state(r.state_seq.load(std::memory_order_acquire)) != REQ_ACTIVE)
+      continue;
+    // active but all bytes received (awaiting test()/H2D) is not pending
+    if (!r.socket_complete()) return true;
+  }
+  return false;
 }
 
 // rx.hdr is complete: locate/validate the posted request (see transport.h).

@@ -264,6 +264,10 @@ SendRequest* claim_chunk(SendComm* c, int sock_idx, uint32_t* off,
 enum RxResult { RX_WAIT = 0, RX_PARKED = 1, RX_CLOSED = 2 };
 // Greedy nonblocking drain of the rx state machine (see transport.cc).
 int drain_recv(TcpSock* s);
+// Any posted recv still waiting for socket bytes?  (A fully-received
+// request that merely awaits test()/H2D drain does NOT make peer EOF an
+// error.)
+bool recv_socket_incomplete(RecvComm* c);
 // rx.hdr is complete: locate/validate the posted request and prime
 // rx.target/remaining.  Returns 0 = proceed, 1 = park (request not yet
 // posted), -1 = protocol error (comm error set).

@@ -353,14 +353,9 @@ class UringIoThread : public IIoThread {
       return;
     }
     if (res == 0) {  // EOF
-      if (!s->rx.in_payload && s->rx.hdr_got == 0) {
-        bool pending = false;
-        for (auto& r : c->reqs)
-          if (ss_state(r.state_seq.load(std::memory_order_acquire)) ==
-              REQ_ACTIVE)
-            pending = true;
-        if (!pending) return;  // orderly shutdown: just stop re-arming
-      }
+      if (!s->rx.in_payload && s->rx.hdr_got == 0 &&
+          !recv_socket_incomplete(c))
+        return;  // orderly shutdown: just stop re-arming
       c->error.store(ECONNRESET);
       BNET_WARN("bnet(uring) recv eof mid-protocol");
       return;

@@ -264,3 +264,122 @@ def test_malformed_frame_errors_not_hangs():
     assert q.get(timeout=120) == "errored"
     proc.join(30)
     assert proc.exitcode == 0
+
+
+def _iso_receiver(env, q, ctl):
+    for k, v in env.items():
+        os.environ[k] = v
+    import ctypes as C
+    import sys
+    import time
+
+    sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+    from baguanet.plugin import Plugin
+
+    p = Plugin()
+    # two listeners: one for the doomed sender, one for the healthy one
+    h1, l1 = p.listen(0)
+    h2, l2 = p.listen(0)
+    q.put((bytes(h1), bytes(h2)))
+    r1 = r2 = None
+    t0 = time.monotonic()
+    while (r1 is None or r2 is None) and time.monotonic() - t0 < 60:
+        if r1 is None:
+            r1 = p.accept(l1)
+        if r2 is None:
+            r2 = p.accept(l2)
+    assert r1 is not None and r2 is not None
+    mh = p.reg_mr(r2, None, 0)
+    ctl.recv()  # wait until the doomed sender is dead
+    # post the recv only now: the kernel buffered a few MB of the 256 MB
+    # message before the sender died, so EOF lands mid-message
+    doomed_buf = C.create_string_buffer(256 << 20)
+    dmh = p.reg_mr(r1, doomed_buf, 256 << 20)
+    doomed = p.irecv(r1, doomed_buf, 256 << 20, dmh)
+    # the healthy comm must still move data correctly
+    for i in range(5):
+        buf = C.create_string_buffer(100000 + 1)
+        req = None
+        while req is None:
+            req = p.irecv(r2, buf, 100000, mh)
+        assert p.wait(req, 60) == 100000
+        assert buf.raw[:4] == bytes([i] * 4)
+    # and the doomed comm must surface its error
+    t0 = time.monotonic()
+    while time.monotonic() - t0 < 60:
+        try:
+            done, _ = p.test(doomed)
+        except RuntimeError:
+            q.put("ok")
+            return
+        assert not done
+        time.sleep(0.01)
+    q.put("doomed-no-error")
+
+
+def _iso_doomed_sender(env, handle_bytes, q):
+    for k, v in env.items():
+        os.environ[k] = v
+    import ctypes as C
+    import time
+
+    from baguanet.plugin import Plugin
+
+    p = Plugin()
+    handle = (C.c_char * len(handle_bytes)).from_buffer_copy(handle_bytes)
+    scomm = None
+    while scomm is None:
+        scomm = p.connect(0, handle)
+    buf = C.create_string_buffer(256 << 20)
+    mh = p.reg_mr(scomm, buf, 256 << 20)
+    p.isend(scomm, buf, 256 << 20, mh)
+    q.put("started")
+    time.sleep(0.3)
+    os._exit(42)
+
+
+def _iso_healthy_sender(env, handle_bytes, q):
+    for k, v in env.items():
+        os.environ[k] = v
+    import ctypes as C
+
+    from baguanet.plugin import Plugin
+
+    p = Plugin()
+    handle = (C.c_char * len(handle_bytes)).from_buffer_copy(handle_bytes)
+    scomm = None
+    while scomm is None:
+        scomm = p.connect(0, handle)
+    mh = p.reg_mr(scomm, None, 0)
+    for i in range(5):
+        payload = bytes([i] * 4) + os.urandom(100000 - 4)
+        buf = C.create_string_buffer(payload, 100000)
+        req = None
+        while req is None:
+            req = p.isend(scomm, buf, 100000, mh)
+        assert p.wait(req, 60) == 100000
+    q.put("healthy-done")
+    p.close_send(scomm)
+
+
+def test_error_isolation_between_comms():
+    """A dead peer errors ONLY its own comm; other comms keep flowing."""
+    env = {"NCCL_SOCKET_IFNAME": "lo", "BNET_NSTREAMS": "2"}
+    ctx = mp.get_context("spawn")
+    qr, qd, qh = ctx.Queue(), ctx.Queue(), ctx.Queue()
+    ca, cb = ctx.Pipe()
+    pr = ctx.Process(target=_iso_receiver, args=(env, qr, ca))
+    pr.start()
+    h1, h2 = qr.get(timeout=60)
+    pd = ctx.Process(target=_iso_doomed_sender, args=(env, h1, qd))
+    ph = ctx.Process(target=_iso_healthy_sender, args=(env, h2, qh))
+    pd.start()
+    ph.start()
+    assert qd.get(timeout=60) == "started"
+    pd.join(30)
+    cb.send("doomed-dead")
+    assert qh.get(timeout=90) == "healthy-done"
+    assert qr.get(timeout=120) == "ok"
+    pr.join(30)
+    ph.join(30)
+    assert pr.exitcode == 0
