@@ -18,14 +18,20 @@ sys.path.insert(0, os.path.join(os.path.dirname(os.path.dirname(
     os.path.abspath(__file__))), "tests"))
 
 
-def _dump_stall(inflight, rreq, sreq):
+def _dump_stall(inflight, rreq, sreq, plugin=None, rcomm=None):
     """A refused post with depth < 32 means some message stalled — dump
     each in-flight entry's completion state before failing."""
+    import ctypes as C
+
     print(f"POST REFUSED: rreq={rreq} sreq={sreq}; "
           f"{len(inflight)} in flight:")
     for i, it in enumerate(inflight):
         print(f"  [{i}] nbytes={it['nbytes']} sdone={it['sdone']} "
               f"rdone={it['rdone']}")
+    if plugin is not None and rcomm is not None:
+        buf = C.create_string_buffer(8192)
+        plugin.lib.bnet_dump_recv_state(rcomm, buf, 8192)
+        print("recv comm state:", buf.value.decode())
     raise AssertionError("post refused — stalled message (state above)")
 
 
@@ -83,7 +89,7 @@ def main():
                 sreq = p.isend(scomm, C.c_void_p(src.data_ptr()), nbytes,
                                smh)
                 if rreq is None or sreq is None:
-                    _dump_stall(inflight, rreq, sreq)
+                    _dump_stall(inflight, rreq, sreq, p, rcomm)
 
                 def verify(src=src, dst=dst):
                     torch.cuda.synchronize()
@@ -99,7 +105,7 @@ def main():
                 rreq = p.irecv(rcomm, rbuf, size, rmh_h)
                 sreq = p.isend(scomm, sbuf, size, smh_h)
                 if rreq is None or sreq is None:
-                    _dump_stall(inflight, rreq, sreq)
+                    _dump_stall(inflight, rreq, sreq, p, rcomm)
 
                 def verify(payload=payload, rbuf=rbuf, sbuf=sbuf, size=size):
                     assert rbuf.raw[:size] == payload, "payload corrupt"

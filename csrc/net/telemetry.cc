@@ -6,6 +6,7 @@
 #include <cstdlib>
 
 #include "baguanet/config.h"
+#include "transport.h"
 
 namespace baguanet {
 
@@ -134,6 +135,37 @@ __attribute__((visibility("default"))) int bnet_config_json(char* buf,
                   c.nstreams, c.min_chunk, c.max_chunk, c.io_threads,
                   c.sockbuf, (int)c.cuda_ptr, c.stage_pool, c.stage_chunk,
                   c.stage_kernel, c.backlog, c.implement.c_str());
+}
+
+// Debug: dump a recv comm's request slots + socket rx state (stall
+// triage from test harnesses).
+__attribute__((visibility("default"))) int bnet_dump_recv_state(
+    void* recv_comm, char* buf, int len) {
+  using namespace baguanet;
+  auto* c = (RecvComm*)recv_comm;
+  int off = 0;
+  off += snprintf(buf + off, len - off, "slots:");
+  for (int i = 0; i < NCCL_NET_MAX_REQUESTS; i++) {
+    auto& r = c->reqs[i];
+    uint64_t ss = r.state_seq.load();
+    if (ss_state(ss) != REQ_ACTIVE) continue;
+    off += snprintf(buf + off, len - off,
+                    " [%d seq=%u total=%ld recvd=%u stage=%d done_sock=%d]",
+                    i, ss_seq(ss), (long)r.total.load(), r.received.load(),
+                    r.stage ? 1 : 0, r.socket_complete() ? 1 : 0);
+    if (off >= len - 128) break;
+  }
+  off += snprintf(buf + off, len - off, " socks:");
+  for (auto* s : c->socks) {
+    off += snprintf(buf + off, len - off,
+                    " {idx=%d parked=%d inpay=%d hdrgot=%u rem=%u seq=%u "
+                    "op=%d epollin_off=%d}",
+                    s->idx, (int)s->parked.load(), (int)s->rx.in_payload,
+                    s->rx.hdr_got, s->rx.remaining, s->rx.hdr.seq,
+                    (int)s->ur.op, 0);
+    if (off >= len - 160) break;
+  }
+  return off;
 }
 
 __attribute__((visibility("default"))) void bnet_dump_metrics(const char* path) {

@@ -60,4 +60,5 @@ extern "C" {
 void bnet_dump_metrics(const char* path);
 void bnet_dump_trace(const char* path);
 int bnet_config_json(char* buf, int len);
+int bnet_dump_recv_state(void* recv_comm, char* buf, int len);
 }
