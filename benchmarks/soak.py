@@ -18,7 +18,7 @@ sys.path.insert(0, os.path.join(os.path.dirname(os.path.dirname(
     os.path.abspath(__file__))), "tests"))
 
 
-def _dump_stall(inflight, rreq, sreq, plugin=None, rcomm=None):
+def _dump_stall(inflight, rreq, sreq, plugin=None, rcomm=None, scomm=None):
     """A refused post with depth < 32 means some message stalled — dump
     each in-flight entry's completion state before failing."""
     import ctypes as C
@@ -32,6 +32,10 @@ def _dump_stall(inflight, rreq, sreq, plugin=None, rcomm=None):
         buf = C.create_string_buffer(8192)
         plugin.lib.bnet_dump_recv_state(rcomm, buf, 8192)
         print("recv comm state:", buf.value.decode())
+    if plugin is not None and scomm is not None:
+        buf = C.create_string_buffer(8192)
+        plugin.lib.bnet_dump_send_state(scomm, buf, 8192)
+        print("send comm state:", buf.value.decode())
     raise AssertionError("post refused — stalled message (state above)")
 
 
@@ -89,7 +93,7 @@ def main():
                 sreq = p.isend(scomm, C.c_void_p(src.data_ptr()), nbytes,
                                smh)
                 if rreq is None or sreq is None:
-                    _dump_stall(inflight, rreq, sreq, p, rcomm)
+                    _dump_stall(inflight, rreq, sreq, p, rcomm, scomm)
 
                 def verify(src=src, dst=dst):
                     torch.cuda.synchronize()
@@ -105,7 +109,7 @@ def main():
                 rreq = p.irecv(rcomm, rbuf, size, rmh_h)
                 sreq = p.isend(scomm, sbuf, size, smh_h)
                 if rreq is None or sreq is None:
-                    _dump_stall(inflight, rreq, sreq, p, rcomm)
+                    _dump_stall(inflight, rreq, sreq, p, rcomm, scomm)
 
                 def verify(payload=payload, rbuf=rbuf, sbuf=sbuf, size=size):
                     assert rbuf.raw[:size] == payload, "payload corrupt"

@@ -159,10 +159,40 @@ __attribute__((visibility("default"))) int bnet_dump_recv_state(
   for (auto* s : c->socks) {
     off += snprintf(buf + off, len - off,
                     " {idx=%d parked=%d inpay=%d hdrgot=%u rem=%u seq=%u "
-                    "op=%d epollin_off=%d}",
+                    "op=%d epollin=%d}",
                     s->idx, (int)s->parked.load(), (int)s->rx.in_payload,
                     s->rx.hdr_got, s->rx.remaining, s->rx.hdr.seq,
-                    (int)s->ur.op, 0);
+                    (int)s->ur.op, (int)s->epollin_on);
+    if (off >= len - 160) break;
+  }
+  return off;
+}
+
+__attribute__((visibility("default"))) int bnet_dump_send_state(
+    void* send_comm, char* buf, int len) {
+  using namespace baguanet;
+  auto* c = (SendComm*)send_comm;
+  int off = 0;
+  off += snprintf(buf + off, len - off, "oldest=%u next=%u slots:",
+                  c->oldest.load(), c->seq_next);
+  for (int i = 0; i < NCCL_NET_MAX_REQUESTS; i++) {
+    auto& r = c->reqs[i];
+    uint64_t ss = r.state_seq.load();
+    if (ss_state(ss) != REQ_ACTIVE) continue;
+    off += snprintf(buf + off, len - off,
+                    " [%d seq=%u total=%u cursor=%u avail=%u sent=%u]",
+                    i, ss_seq(ss), r.total, r.cursor.load(), r.avail.load(),
+                    r.sent.load());
+    if (off >= len - 128) break;
+  }
+  off += snprintf(buf + off, len - off, " socks:");
+  for (auto* s : c->socks) {
+    off += snprintf(buf + off, len - off,
+                    " {idx=%d txact=%d txdone=%u txlen=%u wantout=%d "
+                    "sndidle=%d op=%d nch=%d}",
+                    s->idx, (int)s->tx.active, s->tx.done, s->tx.hdr.len,
+                    (int)s->want_epollout, (int)s->snd_idle.load(),
+                    (int)s->ur.op, s->ur.nchunks);
     if (off >= len - 160) break;
   }
   return off;

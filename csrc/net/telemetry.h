@@ -61,4 +61,5 @@ void bnet_dump_metrics(const char* path);
 void bnet_dump_trace(const char* path);
 int bnet_config_json(char* buf, int len);
 int bnet_dump_recv_state(void* recv_comm, char* buf, int len);
+int bnet_dump_send_state(void* send_comm, char* buf, int len);
 }

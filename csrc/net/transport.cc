@@ -105,6 +105,7 @@ void IoThread::handle_tasks() {
         epoll_event ev{};
         ev.events = s->is_recv ? EPOLLIN : 0;
         ev.data.ptr = s;
+        s->epollin_on = s->is_recv;
         epoll_ctl(epfd_, EPOLL_CTL_ADD, s->fd, &ev);
         progress(s);
         break;
@@ -424,6 +425,7 @@ void IoThread::progress_recv(TcpSock* s) {
       epoll_event ev{};  // resumed from parked: re-arm EPOLLIN
       ev.events = EPOLLIN;
       ev.data.ptr = s;
+      s->epollin_on = true;
       epoll_ctl(epfd_, EPOLL_CTL_MOD, s->fd, &ev);
     }
     return;
@@ -434,6 +436,7 @@ void IoThread::progress_recv(TcpSock* s) {
     epoll_event ev{};
     ev.events = 0;
     ev.data.ptr = s;
+    s->epollin_on = false;
     epoll_ctl(epfd_, EPOLL_CTL_MOD, s->fd, &ev);
   }
 }

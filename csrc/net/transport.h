@@ -148,6 +148,7 @@ struct TcpSock {
   int idx = 0;  // position within its comm's socket list
   bool is_recv = false;
   bool want_epollout = false;
+  bool epollin_on = false;  // diagnostic mirror of the epoll interest
   // send-side idle flag for the kick fast path: set (seq_cst) by the IO
   // thread right before its final failed claim re-check; isend publishes
   // the request (seq_cst + fence) then kicks only sockets with snd_idle
