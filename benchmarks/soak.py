@@ -39,6 +39,11 @@ def _dump_stall(inflight, rreq, sreq, plugin=None, rcomm=None, scomm=None):
     raise AssertionError("post refused — stalled message (state above)")
 
 
+_SIZES = [0, 64, 5000, 65536, 300_000, 1 << 20, 3 << 20]
+if os.environ.get("BNET_SOAK_SIZES"):
+    _SIZES = [int(x) for x in os.environ["BNET_SOAK_SIZES"].split(",")]
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--seconds", type=float, default=30.0)
@@ -71,8 +76,7 @@ def main():
     while time.monotonic() - t0 < args.seconds or inflight:
         while (len(inflight) < 12
                and time.monotonic() - t0 < args.seconds):
-            size = rng.choice(
-                [0, 64, 5000, 65536, 300_000, 1 << 20, 3 << 20])
+            size = rng.choice(_SIZES)
             use_gpu = args.gpu and torch is not None and rng.random() < 0.5
             if use_gpu and size > 0:
                 n = max(size // 4, 1)
