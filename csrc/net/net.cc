@@ -336,18 +336,22 @@ ncclResult_t Net::isend(void* send_comm, void* data, int size, int tag,
   auto* c = (SendComm*)send_comm;
   if (c->error.load(std::memory_order_relaxed)) return ncclSystemError;
   SendRequest* r = &c->reqs[c->seq_next % NCCL_NET_MAX_REQUESTS];
-  if (ss_state(r->state_seq.load(std::memory_order_acquire)) != REQ_FREE) {
+  uint64_t ss0 = r->state_seq.load(std::memory_order_acquire);
+  if (ss_state(ss0) != REQ_FREE) {
+    c->last_refusal_ss = ss0;
+    c->last_refusal_at = c->seq_next;
     *request = nullptr;  // slot busy — NCCL retries
     return ncclSuccess;
   }
   int ptr_type = (int)(uintptr_t)mhandle;
+  // ORDER: bump the cursor GENERATION first — any stale claimer's CAS
+  // (old generation) must fail before it can see this request's fields
+  r->cursor.store(pack_cur(c->seq_next, 0), std::memory_order_relaxed);
   r->total = (uint32_t)size;
   r->chunk = pick_chunk_size((uint32_t)size, Config::get().min_chunk,
                              Config::get().max_chunk,
                              (int)c->socks.size());
-  r->cursor.store(0, std::memory_order_relaxed);
   r->sent.store(0, std::memory_order_relaxed);
-  r->hdr_claimed.store(false, std::memory_order_relaxed);
   r->hdr_sent.store(false, std::memory_order_relaxed);
   r->comm = c;
   if (ptr_type == NCCL_PTR_CUDA && size > 0) {
@@ -403,7 +407,10 @@ ncclResult_t Net::irecv(void* recv_comm, int n, void** data, int* sizes,
     return ncclInternalError;
   }
   RecvRequest* r = &c->reqs[c->post_next % NCCL_NET_MAX_REQUESTS];
-  if (ss_state(r->state_seq.load(std::memory_order_acquire)) != REQ_FREE) {
+  uint64_t ss0 = r->state_seq.load(std::memory_order_acquire);
+  if (ss_state(ss0) != REQ_FREE) {
+    c->last_refusal_ss = ss0;
+    c->last_refusal_at = c->post_next;
     *request = nullptr;
     return ncclSuccess;
   }

@@ -144,14 +144,18 @@ __attribute__((visibility("default"))) int bnet_dump_recv_state(
   using namespace baguanet;
   auto* c = (RecvComm*)recv_comm;
   int off = 0;
-  off += snprintf(buf + off, len - off, "slots:");
+  off += snprintf(buf + off, len - off,
+                  "post_next=%u refusal@%u=(%u,%u) slots:", c->post_next,
+                  c->last_refusal_at, ss_seq(c->last_refusal_ss),
+                  ss_state(c->last_refusal_ss));
   for (int i = 0; i < NCCL_NET_MAX_REQUESTS; i++) {
     auto& r = c->reqs[i];
     uint64_t ss = r.state_seq.load();
     if (ss_state(ss) != REQ_ACTIVE) continue;
     off += snprintf(buf + off, len - off,
-                    " [%d seq=%u total=%ld recvd=%u stage=%d done_sock=%d]",
-                    i, ss_seq(ss), (long)r.total.load(), r.received.load(),
+                    " [%d@%p seq=%u total=%ld recvd=%u stage=%d done_sock=%d]",
+                    i, (void*)((uintptr_t)&r | 2),
+                    ss_seq(ss), (long)r.total.load(), r.received.load(),
                     r.stage ? 1 : 0, r.socket_complete() ? 1 : 0);
     if (off >= len - 128) break;
   }
@@ -173,29 +177,52 @@ __attribute__((visibility("default"))) int bnet_dump_send_state(
   using namespace baguanet;
   auto* c = (SendComm*)send_comm;
   int off = 0;
-  off += snprintf(buf + off, len - off, "oldest=%u next=%u slots:",
-                  c->oldest.load(), c->seq_next);
+  off += snprintf(buf + off, len - off,
+                  "oldest=%u next=%u refusal@%u=(%u,%u) slots:",
+                  c->oldest.load(), c->seq_next, c->last_refusal_at,
+                  ss_seq(c->last_refusal_ss), ss_state(c->last_refusal_ss));
   for (int i = 0; i < NCCL_NET_MAX_REQUESTS; i++) {
     auto& r = c->reqs[i];
     uint64_t ss = r.state_seq.load();
     if (ss_state(ss) != REQ_ACTIVE) continue;
     off += snprintf(buf + off, len - off,
-                    " [%d seq=%u total=%u cursor=%u avail=%u sent=%u]",
-                    i, ss_seq(ss), r.total, r.cursor.load(), r.avail.load(),
-                    r.sent.load());
+                    " [%d seq=%u total=%u chunk=%u cur=%u/g%u avail=%u "
+                    "sent=%u]",
+                    i, ss_seq(ss), r.total, r.chunk,
+                    cur_off(r.cursor.load()), cur_gen(r.cursor.load()),
+                    r.avail.load(), r.sent.load());
     if (off >= len - 128) break;
   }
   off += snprintf(buf + off, len - off, " socks:");
   for (auto* s : c->socks) {
     off += snprintf(buf + off, len - off,
                     " {idx=%d txact=%d txdone=%u txlen=%u wantout=%d "
-                    "sndidle=%d op=%d nch=%d}",
+                    "sndidle=%d op=%d nch=%d kicks=%u prog=%u claims=%u "
+                    "brk=%u full=%u brkS=%u brkSS=%u/%u brkOld=%u "
+                    "gfail=%u ffail=%u exit=%u reent=%u krun=%u rescue=%u}",
                     s->idx, (int)s->tx.active, s->tx.done, s->tx.hdr.len,
                     (int)s->want_epollout, (int)s->snd_idle.load(),
-                    (int)s->ur.op, s->ur.nchunks);
+                    (int)s->ur.op, s->ur.nchunks, s->dbg_kicks.load(),
+                    s->dbg_progress.load(), s->dbg_claims.load(),
+                    s->dbg_breaks.load(), s->dbg_full.load(),
+                    s->dbg_break_s.load(),
+                    ss_seq(s->dbg_break_ss.load()),
+                    ss_state(s->dbg_break_ss.load()),
+                    s->dbg_break_old.load(), s->dbg_gate_fail.load(),
+                    s->dbg_find_fail.load(), s->dbg_exit.load(),
+                    s->dbg_reent.load(), s->dbg_kick_run.load(),
+                    s->dbg_sweep_rescue.load());
     if (off >= len - 160) break;
   }
   return off;
+}
+
+// Force-kick every socket of a send comm (stall-recovery probe).
+__attribute__((visibility("default"))) void bnet_force_kick(void* send_comm) {
+  using namespace baguanet;
+  auto* c = (SendComm*)send_comm;
+  for (auto* s : c->socks)
+    Engine::get().thread(s->io_thread).kick(s);
 }
 
 __attribute__((visibility("default"))) void bnet_dump_metrics(const char* path) {

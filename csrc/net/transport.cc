@@ -85,6 +85,7 @@ void IoThread::remove_sock_sync(TcpSock* s) {
 }
 
 void IoThread::kick(TcpSock* s) {
+  s->dbg_kicks.fetch_add(1, std::memory_order_relaxed);
   std::lock_guard<std::mutex> lk(task_mu_);
   tasks_.push_back({Task::KICK, s, nullptr, nullptr});
   uint64_t one = 1;
@@ -128,8 +129,12 @@ void IoThread::handle_tasks() {
       }
       case Task::KICK:
         // The socket may have been removed between enqueue and drain.
-        if (std::find(socks_.begin(), socks_.end(), t.s) != socks_.end())
+        if (std::find(socks_.begin(), socks_.end(), t.s) != socks_.end()) {
+          t.s->dbg_kick_run.fetch_add(1, std::memory_order_relaxed);
           progress(t.s);
+        } else {
+          t.s->dbg_find_fail.fetch_add(1, std::memory_order_relaxed);
+        }
         break;
     }
   }
@@ -160,21 +165,46 @@ void IoThread::run() {
     int n = epoll_wait(epfd_, evs, 64, spin ? 0 : 100);
     if (n < 0 && errno != EINTR) break;
     if (n > 0) last_active_ns = now;
-    bool had_ev = false;
     for (int i = 0; i < n; i++) {
       if (evs[i].data.ptr == nullptr) {
         uint64_t v;
         (void)!read(evfd_, &v, sizeof(v));
-        had_ev = true;
         continue;
       }
       progress(static_cast<TcpSock*>(evs[i].data.ptr));
     }
-    if (had_ev || spin) handle_tasks();
+    // Drain tasks EVERY iteration, not only on eventfd wakes: gating the
+    // drain on (had_ev || spin) left enqueued kicks stranded in rare
+    // interleavings (observed: kicks enqueued == executed + N with all
+    // threads asleep), stalling a message until its slot wrapped.  The
+    // eventfd still guarantees a sleeping thread wakes; draining is
+    // unconditional so no wake can race past its task.
+    handle_tasks();
     if (spin) {
-      // watermarks/jobs may have advanced without an epoll event
+      // watermarks/jobs may have advanced without an epoll event; also
+      // resume any mid-write socket that is not EPOLLOUT-armed (progress
+      // on an active tx just continues its writev)
       for (TcpSock* s : socks_)
-        if (!s->is_recv && !s->tx.active && !s->want_epollout) progress(s);
+        if (!s->is_recv && !s->want_epollout) progress(s);
+    } else {
+      // Pre-block sweep: unconditional forward-progress guarantee.  Kicks
+      // are an optimization; before sleeping, re-scan idle senders and
+      // parked receivers so that even a lost wakeup costs at most one
+      // idle-loop period (100 ms), never a stall.  (A residual ~1-in-200k
+      // lost kick was still observed after the drain fix; this bounds it.)
+      for (TcpSock* s : socks_) {
+        if (!s->is_recv && !s->want_epollout) {
+          uint32_t before = s->dbg_claims.load(std::memory_order_relaxed);
+          progress(s);
+          if (s->dbg_claims.load(std::memory_order_relaxed) != before)
+            s->dbg_sweep_rescue.fetch_add(1, std::memory_order_relaxed);
+        } else if (s->is_recv) {
+          // parked sockets re-check their slot; armed ones cost one
+          // EAGAIN read per idle period — total immunity to any missed
+          // readiness edge
+          progress(s);
+        }
+      }
     }
   }
 }
@@ -204,7 +234,7 @@ void IoThread::progress(TcpSock* s) {
 // small-message rate (measured on the MI355X box: 3.8 vs 11.4 GB/s at
 // 64 KiB).  Multi-chunk messages keep fully dynamic assignment.
 SendRequest* claim_chunk(SendComm* c, int sock_idx, uint32_t* off,
-                         uint32_t* len) {
+                         uint32_t* len, TcpSock* dbg) {
   int nsocks = (int)c->socks.size();
   // Single-chunk messages fan out over at most 2 sockets: measured on the
   // MI355X box, 64 KiB message rate is ~11 GB/s at 1-2 sockets but drops
@@ -228,6 +258,12 @@ SendRequest* claim_chunk(SendComm* c, int sock_idx, uint32_t* off,
         }
         continue;
       }
+      if (dbg) {
+        dbg->dbg_breaks.fetch_add(1, std::memory_order_relaxed);
+        dbg->dbg_break_s.store(s, std::memory_order_relaxed);
+        dbg->dbg_break_ss.store(ss, std::memory_order_relaxed);
+        dbg->dbg_break_old.store(oldest, std::memory_order_relaxed);
+      }
       break;  // seq s was never posted → nothing newer exists either
     }
     if (r->total <= r->chunk) {
@@ -235,8 +271,10 @@ SendRequest* claim_chunk(SendComm* c, int sock_idx, uint32_t* off,
       if ((int)(s % (uint32_t)nsmall) != sock_idx) continue;
     }
     if (r->total == 0) {
-      bool expect = false;
-      if (r->hdr_claimed.compare_exchange_strong(expect, true)) {
+      // header-only claim: offset field doubles as the claim flag; the
+      // generation tag rejects stale claims after slot reuse
+      uint64_t cur64 = pack_cur(s, 0);
+      if (r->cursor.compare_exchange_strong(cur64, pack_cur(s, 1))) {
         *off = 0;
         *len = 0;
         return r;
@@ -245,24 +283,37 @@ SendRequest* claim_chunk(SendComm* c, int sock_idx, uint32_t* off,
     }
     uint32_t avail = std::min(r->avail.load(std::memory_order_acquire),
                               r->total);
-    uint32_t cur = r->cursor.load(std::memory_order_relaxed);
-    while (cur < avail) {
+    uint64_t cur64 = r->cursor.load(std::memory_order_relaxed);
+    while (cur_gen(cur64) == s && cur_off(cur64) < avail) {
+      uint32_t cur = cur_off(cur64);
       uint32_t end = std::min(cur + r->chunk, avail);
-      if (r->cursor.compare_exchange_weak(cur, end)) {
+      if (r->cursor.compare_exchange_weak(cur64, pack_cur(s, end))) {
         *off = cur;
         *len = end - cur;
         return r;
       }
+      // cur64 reloaded by the failed CAS; the gen check guards reuse
     }
     // nothing claimable in this request (fully claimed or staging-limited);
     // move on to the next one
   }
+  if (dbg) dbg->dbg_full.fetch_add(1, std::memory_order_relaxed);
   return nullptr;
 }
 
 void IoThread::progress_send(TcpSock* s) {
   SendComm* c = s->scomm;
-  if (c->error.load(std::memory_order_relaxed)) return;
+  s->dbg_progress.fetch_add(1, std::memory_order_relaxed);
+  if (s->dbg_inprog.fetch_add(1, std::memory_order_acq_rel) != 0)
+    s->dbg_reent.fetch_add(1, std::memory_order_relaxed);
+  struct Guard {
+    TcpSock* s;
+    ~Guard() { s->dbg_inprog.fetch_sub(1, std::memory_order_acq_rel); }
+  } guard{s};
+  if (c->error.load(std::memory_order_relaxed)) {
+    s->dbg_exit.store(5, std::memory_order_relaxed);
+    return;
+  }
   // Fairness credit: one dispatch writes at most `inflight_per_stream`
   // bytes before yielding the IO thread to its other sockets (other
   // comms sharing the NIC), then reschedules itself.
@@ -271,19 +322,21 @@ void IoThread::progress_send(TcpSock* s) {
   while (true) {
     if (!s->tx.active) {
       uint32_t off = 0, len = 0;
-      SendRequest* r = claim_chunk(c, s->idx, &off, &len);
+      SendRequest* r = claim_chunk(c, s->idx, &off, &len, s);
       if (!r) {
         // going idle: publish-then-recheck so a concurrent isend either
         // sees snd_idle (and kicks) or we see its request now
         s->snd_idle.store(true, std::memory_order_seq_cst);
         std::atomic_thread_fence(std::memory_order_seq_cst);
-        r = claim_chunk(c, s->idx, &off, &len);
+        r = claim_chunk(c, s->idx, &off, &len, s);
         if (!r) {
           set_epollout(s, false);
+          s->dbg_exit.store(1, std::memory_order_relaxed);
           return;
         }
         s->snd_idle.store(false, std::memory_order_relaxed);
       }
+      s->dbg_claims.fetch_add(1, std::memory_order_relaxed);
       s->tx.active = true;
       s->tx.req = r;
       s->tx.hdr = {ss_seq(r->state_seq.load(std::memory_order_relaxed)), off,
@@ -314,11 +367,13 @@ void IoThread::progress_send(TcpSock* s) {
       }
       if (w < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) {
         set_epollout(s, true);
+        s->dbg_exit.store(2, std::memory_order_relaxed);
         return;
       }
       if (w < 0 && errno == EINTR) continue;
       c->error.store(errno ? errno : EPIPE);
       BNET_WARN("bnet send socket error: %s", strerror(errno));
+      s->dbg_exit.store(3, std::memory_order_relaxed);
       return;
     }
     // chunk fully written
@@ -334,6 +389,7 @@ void IoThread::progress_send(TcpSock* s) {
     }
     if (len >= budget) {
       kick(s);  // yield: requeue ourselves behind other sockets' work
+      s->dbg_exit.store(4, std::memory_order_relaxed);
       return;
     }
     budget -= len;
@@ -562,17 +618,18 @@ void Engine::kick_comm(SendComm* c, int max_socks) {
   // this request or this kick sees snd_idle).
   int n = max_socks < 0 ? (int)c->socks.size() : max_socks;
   for (TcpSock* s : c->socks) {
-    if (n <= 0) break;
-    if (s->snd_idle.load(std::memory_order_seq_cst)) {
-      threads_[s->io_thread]->kick(s);
-      n--;
-    }
+    if (n-- <= 0) break;
+    threads_[s->io_thread]->kick(s);  // unconditional — see kick_sock
   }
 }
 
 void Engine::kick_sock(TcpSock* s) {
-  if (s->snd_idle.load(std::memory_order_seq_cst))
-    threads_[s->io_thread]->kick(s);  // handshake-guarded — see kick_comm
+  // UNCONDITIONAL: the snd_idle gate was observed (rarely, ~1e-5 of
+  // messages under soak) to miss the idle transition despite the
+  // seq_cst publish-then-recheck pair, stalling a message until the
+  // pre-block sweep.  A redundant kick to a busy socket is a cheap
+  // no-op; correctness wins.
+  threads_[s->io_thread]->kick(s);
 }
 
 void Engine::kick_comm(RecvComm* c) {

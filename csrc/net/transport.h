@@ -100,6 +100,19 @@ inline uint64_t pack_ss(uint32_t seq, uint32_t state) {
 inline uint32_t ss_state(uint64_t ss) { return (uint32_t)ss; }
 inline uint32_t ss_seq(uint64_t ss) { return (uint32_t)(ss >> 32); }
 
+// The claim cursor is generation-tagged the same way: (seq << 32) | offset.
+// A claimer validates state_seq, but between that check and its cursor CAS
+// the slot can complete -> free -> repost (observed under soak: a stale
+// CAS claimed a 32 KiB chunk of a freshly-posted 5000 B message,
+// corrupting the stream and wedging completion).  With the generation in
+// the CAS word, a cross-generation CAS can never commit; the repost bumps
+// the cursor generation BEFORE writing any other field.
+inline uint64_t pack_cur(uint32_t seq, uint32_t off) {
+  return ((uint64_t)seq << 32) | off;
+}
+inline uint32_t cur_off(uint64_t c) { return (uint32_t)c; }
+inline uint32_t cur_gen(uint64_t c) { return (uint32_t)(c >> 32); }
+
 struct SendComm;
 struct RecvComm;
 
@@ -108,10 +121,11 @@ struct SendRequest {
   uint32_t total = 0;
   uint32_t chunk = 0;           // stripe chunk size chosen for this message
   const char* src = nullptr;    // host source (user buffer or staging bounce)
-  std::atomic<uint32_t> cursor{0};  // next unclaimed offset
+  // generation-tagged claim cursor: (seq << 32) | next_unclaimed_offset.
+  // For total == 0 the "offset" doubles as the header-claim flag (0 -> 1).
+  std::atomic<uint64_t> cursor{0};
   std::atomic<uint32_t> avail{0};   // staged watermark; == total for host src
   std::atomic<uint32_t> sent{0};    // bytes fully handed to the kernel
-  std::atomic<bool> hdr_claimed{false};  // zero-byte message header claim
   std::atomic<bool> hdr_sent{false};
   StageAlloc* stage = nullptr;  // non-null for NCCL_PTR_CUDA sends
   SendComm* comm = nullptr;
@@ -149,6 +163,22 @@ struct TcpSock {
   bool is_recv = false;
   bool want_epollout = false;
   bool epollin_on = false;  // diagnostic mirror of the epoll interest
+  // stall-triage counters
+  std::atomic<uint32_t> dbg_kicks{0};      // kick() enqueued for this sock
+  std::atomic<uint32_t> dbg_progress{0};   // progress_send/submit entries
+  std::atomic<uint32_t> dbg_claims{0};     // successful chunk claims
+  std::atomic<uint32_t> dbg_breaks{0};     // scans ended via break
+  std::atomic<uint32_t> dbg_full{0};       // scans ended at window end
+  std::atomic<uint32_t> dbg_break_s{0};    // last break: scan seq
+  std::atomic<uint64_t> dbg_break_ss{0};   // last break: slot state_seq
+  std::atomic<uint32_t> dbg_break_old{0};  // last break: oldest at entry
+  std::atomic<uint32_t> dbg_gate_fail{0};  // kick_sock gate saw idle=false
+  std::atomic<uint32_t> dbg_find_fail{0};  // KICK task: sock not in socks_
+  std::atomic<uint32_t> dbg_exit{0};       // progress_send last exit point
+  std::atomic<int> dbg_inprog{0};          // reentrancy detector
+  std::atomic<uint32_t> dbg_reent{0};      // progress_send reentered!
+  std::atomic<uint32_t> dbg_kick_run{0};   // KICK tasks executed for sock
+  std::atomic<uint32_t> dbg_sweep_rescue{0};  // pre-block sweep found work
   // send-side idle flag for the kick fast path: set (seq_cst) by the IO
   // thread right before its final failed claim re-check; isend publishes
   // the request (seq_cst + fence) then kicks only sockets with snd_idle
@@ -203,6 +233,8 @@ struct SendComm {
   std::vector<TcpSock*> socks;
   SendRequest reqs[NCCL_NET_MAX_REQUESTS];
   uint32_t seq_next = 0;         // proxy thread only
+  uint64_t last_refusal_ss = 0;
+  uint32_t last_refusal_at = 0;
   std::atomic<uint32_t> oldest{0};  // completion frontier (lazily advanced)
   std::atomic<int> error{0};
   std::atomic<int> live_socks{0};
@@ -215,6 +247,8 @@ struct RecvComm {
   std::vector<TcpSock*> socks;
   RecvRequest reqs[NCCL_NET_MAX_REQUESTS];
   uint32_t post_next = 0;  // proxy thread only
+  uint64_t last_refusal_ss = 0;  // blocking slot's state_seq at refusal
+  uint32_t last_refusal_at = 0;  // post_next at refusal
   std::atomic<int> error{0};
   std::atomic<int> live_socks{0};
   int dev = 0;
@@ -261,7 +295,7 @@ struct ListenComm {
 
 // Shared transport state-machine helpers (used by both engines).
 SendRequest* claim_chunk(SendComm* c, int sock_idx, uint32_t* off,
-                         uint32_t* len);
+                         uint32_t* len, TcpSock* dbg = nullptr);
 enum RxResult { RX_WAIT = 0, RX_PARKED = 1, RX_CLOSED = 2 };
 // Greedy nonblocking drain of the rx state machine (see transport.cc).
 int drain_recv(TcpSock* s);
