@@ -33,6 +33,8 @@ struct StageAlloc {
   char* host = nullptr;
   size_t pool_off = 0;
   uint32_t size = 0;
+  uint64_t t_begin_ns = 0;   // for slow-event diagnostics
+  bool diag_logged = false;
   // send pipeline
   const char* gpu_src = nullptr;
   uint32_t total = 0;
@@ -153,6 +155,7 @@ bool stage_send_begin(StagePool* p, SendRequest* req, const void* src,
   a->total = total;
   a->copy_chunk = Config::get().stage_chunk;
   a->sreq = req;
+  a->t_begin_ns = now_ns();
   req->src = a->host;
   req->stage = a;
   for (uint32_t off = 0; off < total; off += a->copy_chunk) {
@@ -180,8 +183,10 @@ bool stage_poll(StagePool* p) {
     std::lock_guard<std::mutex> lk(p->mu);
     for (auto it = p->inflight.begin(); it != p->inflight.end();) {
       StageAlloc* a = *it;
+      hipError_t qe = hipSuccess;
       while (a->events_done < a->events.size() &&
-             hipEventQuery(a->events[a->events_done]) == hipSuccess) {
+             (qe = hipEventQuery(a->events[a->events_done])) ==
+                 hipSuccess) {
         a->events_done++;
         advanced = true;
         uint32_t avail =
@@ -196,6 +201,26 @@ bool stage_poll(StagePool* p) {
         it = p->inflight.erase(it);
         p->pending.fetch_sub(1, std::memory_order_release);
       } else {
+        // Slow-event diagnostics + one-shot recovery: a D2H copy of a
+        // few hundred KB completes in microseconds; >2 s pending means
+        // the query errors, the stream wedged, or the copy was lost.
+        // Log the exact state once and re-issue the remaining copies.
+        if (!a->diag_logged && now_ns() - a->t_begin_ns > 2'000'000'000ull) {
+          a->diag_logged = true;
+          hipError_t sq = hipStreamQuery(p->d2h);
+          BNET_WARN(
+              "staged D2H stuck >2s: total=%u done=%u/%zu query=%d(%s) "
+              "streamQuery=%d(%s) — re-issuing remaining copies",
+              a->total, a->events_done, a->events.size(), (int)qe,
+              hipGetErrorString(qe), (int)sq, hipGetErrorString(sq));
+          for (uint32_t i = a->events_done; i < a->events.size(); i++) {
+            uint32_t off = i * a->copy_chunk;
+            uint32_t n = std::min(a->copy_chunk, a->total - off);
+            issue_copy(a->host + off, a->gpu_src + off, n,
+                       hipMemcpyDeviceToHost, p->d2h);
+            HIP_WARN(hipEventRecord(a->events[i], p->d2h));
+          }
+        }
         ++it;
       }
     }
