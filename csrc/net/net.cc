@@ -477,6 +477,7 @@ ncclResult_t Net::test(void* request, int* done, int* sizes) {
     }
     if (c->stage_pool && stage_pending(c->stage_pool))
       stage_poll(c->stage_pool);
+    if (r->stage && !r->complete()) stage_send_watchdog(c->stage_pool, r);
     if (r->complete()) {
       *done = 1;
       if (sizes) sizes[0] = (int)r->total;

@@ -238,6 +238,49 @@ bool stage_pending(StagePool* p) {
   return p->pending.load(std::memory_order_acquire) > 0;
 }
 
+void stage_send_watchdog(StagePool* p, SendRequest* req) {
+  StageAlloc* a = (StageAlloc*)req->stage;
+  if (!a) return;
+  std::lock_guard<std::mutex> lk(p->mu);
+  if (req->avail.load(std::memory_order_acquire) >= a->total) return;
+  if (now_ns() - a->t_begin_ns < 2'000'000'000ull) return;
+  bool in_list = false;
+  for (auto* x : p->inflight)
+    if (x == a) in_list = true;
+  hipError_t sq = hipStreamQuery(p->d2h);
+  BNET_WARN(
+      "staged send watchdog: total=%u avail=%u events=%zu done=%u "
+      "in_inflight=%d pool_pending=%d streamQuery=%d(%s) — re-staging",
+      a->total, req->avail.load(), a->events.size(), a->events_done,
+      (int)in_list, p->pending.load(), (int)sq, hipGetErrorString(sq));
+  // re-issue all remaining copies and re-register for polling
+  uint32_t start = a->events_done;
+  for (uint32_t i = start; i < a->events.size(); i++) {
+    uint32_t off = i * a->copy_chunk;
+    uint32_t n = std::min(a->copy_chunk, a->total - off);
+    issue_copy(a->host + off, a->gpu_src + off, n, hipMemcpyDeviceToHost,
+               p->d2h);
+    HIP_WARN(hipEventRecord(a->events[i], p->d2h));
+  }
+  if (a->events.empty()) {
+    // pathological: no events recorded at all — restage from scratch
+    for (uint32_t off = 0; off < a->total; off += a->copy_chunk) {
+      uint32_t n = std::min(a->copy_chunk, a->total - off);
+      issue_copy(a->host + off, a->gpu_src + off, n, hipMemcpyDeviceToHost,
+                 p->d2h);
+      hipEvent_t ev = p->get_event();
+      HIP_WARN(hipEventRecord(ev, p->d2h));
+      a->events.push_back(ev);
+    }
+    a->events_done = 0;
+  }
+  if (!in_list) {
+    p->inflight.push_back(a);
+    p->pending.fetch_add(1, std::memory_order_release);
+  }
+  a->t_begin_ns = now_ns();  // rearm the watchdog
+}
+
 bool stage_recv_begin(StagePool* p, RecvRequest* req, void* dst,
                       uint32_t capacity) {
   std::lock_guard<std::mutex> lk(p->mu);

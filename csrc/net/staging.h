@@ -51,6 +51,12 @@ bool stage_poll(StagePool* p);
 // Any send staging still in flight? (IO threads spin while true.)
 bool stage_pending(StagePool* p);
 
+// Request-level watchdog called from test(): if req's staging has not
+// advanced for >2 s (alloc missing from the poll list, wedged stream,
+// errored event), log the pool/alloc state and re-stage the remaining
+// copies, re-registering the alloc for polling.
+void stage_send_watchdog(StagePool* p, SendRequest* req);
+
 // ---- recv path ----
 // Reserve a pinned bounce of `capacity` bytes for a CUDA recv.  Returns
 // false if pool space is exhausted.
