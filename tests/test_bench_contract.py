@@ -34,6 +34,29 @@ def test_bench_json_contract():
     assert out["config"]["parallelism"] == "dp1"
 
 
+def test_bench_torchrun_world2_cpu():
+    """The driver's SCALE launch shape: torchrun --nproc-per-node N
+    bench.py --gpus N.  On CPU this exercises gloo init, BucketedDDP
+    world=2, the MAX-over-ranks reduction, and the single-JSON-line
+    contract from rank 0 only."""
+    res = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29534",
+         os.path.join(REPO, "bench.py"), "--gpus", "2", "--model",
+         "resnet50", "--steps", "1", "--warmup", "0", "--batch", "2"],
+        capture_output=True, text=True, timeout=600, cwd=REPO,
+    )
+    assert res.returncode == 0, res.stderr[-2000:]
+    lines = [ln for ln in res.stdout.splitlines() if ln.startswith("{")]
+    assert len(lines) == 1, res.stdout  # rank 0 only
+    out = json.loads(lines[0])
+    assert out["n_gpus"] == 2
+    assert out["config"]["parallelism"] == "dp2"
+    assert out["config"]["global_batch"] == 4
+    assert out["value"] > 0
+
+
 def test_allreduce_perf_torchrun_cpu():
     env = dict(os.environ)
     res = subprocess.run(
