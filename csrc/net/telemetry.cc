@@ -188,7 +188,7 @@ __attribute__((visibility("default"))) int bnet_dump_send_state(
     off += snprintf(buf + off, len - off,
                     " [%d seq=%u total=%u chunk=%u cur=%u/g%u avail=%u "
                     "sent=%u]",
-                    i, ss_seq(ss), r.total, r.chunk,
+                    i, ss_seq(ss), r.total.load(), r.chunk.load(),
                     cur_off(r.cursor.load()), cur_gen(r.cursor.load()),
                     r.avail.load(), r.sent.load());
     if (off >= len - 128) break;

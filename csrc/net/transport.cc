@@ -282,11 +282,11 @@ SendRequest* claim_chunk(SendComm* c, int sock_idx, uint32_t* off,
       continue;
     }
     uint32_t avail = std::min(r->avail.load(std::memory_order_acquire),
-                              r->total);
+                              r->total.load(std::memory_order_relaxed));
     uint64_t cur64 = r->cursor.load(std::memory_order_relaxed);
     while (cur_gen(cur64) == s && cur_off(cur64) < avail) {
       uint32_t cur = cur_off(cur64);
-      uint32_t end = std::min(cur + r->chunk, avail);
+      uint32_t end = std::min(cur + r->chunk.load(std::memory_order_relaxed), avail);
       if (r->cursor.compare_exchange_weak(cur64, pack_cur(s, end))) {
         *off = cur;
         *len = end - cur;

@@ -118,8 +118,14 @@ struct RecvComm;
 
 struct SendRequest {
   std::atomic<uint64_t> state_seq{pack_ss(UINT32_MAX, REQ_FREE)};
-  uint32_t total = 0;
-  uint32_t chunk = 0;           // stripe chunk size chosen for this message
+  // total/chunk are atomics ONLY because a stale claimer may read them
+  // while the slot is being reposted for the next generation (the
+  // gen-tagged cursor CAS then rejects the claim, so the value is
+  // discarded — but the concurrent plain read would still be UB; TSan
+  // flags it).  All accesses are relaxed; ordering comes from the
+  // state_seq publish.
+  std::atomic<uint32_t> total{0};
+  std::atomic<uint32_t> chunk{0};  // stripe chunk size for this message
   const char* src = nullptr;    // host source (user buffer or staging bounce)
   // generation-tagged claim cursor: (seq << 32) | next_unclaimed_offset.
   // For total == 0 the "offset" doubles as the header-claim flag (0 -> 1).
@@ -132,8 +138,9 @@ struct SendRequest {
   uint32_t span_slot = UINT32_MAX;
 
   bool complete() const {
-    return total == 0 ? hdr_sent.load(std::memory_order_acquire)
-                      : sent.load(std::memory_order_acquire) == total;
+    uint32_t t = total.load(std::memory_order_relaxed);
+    return t == 0 ? hdr_sent.load(std::memory_order_acquire)
+                  : sent.load(std::memory_order_acquire) == t;
   }
 };
 

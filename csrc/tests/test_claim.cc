@@ -23,7 +23,7 @@ int main() {
       r->total = (q % 4 == 0) ? 5000 : 64;
       r->chunk = 32768;
       r->cursor.store(pack_cur(q, 0));
-      r->avail.store(r->total);
+      r->avail.store(r->total.load());
       r->sent.store(0);
       r->state_seq.store(pack_ss(q, REQ_ACTIVE));
     } else {

@@ -13,7 +13,10 @@
 
 using namespace baguanet;
 
-int main() {
+int main(int argc, char** argv) {
+  // argv[1]: message target (default 30M; TSan runs use a smaller count)
+  const uint64_t kTarget = argc > 1 ? strtoull(argv[1], nullptr, 10)
+                                    : 30'000'000;
   SendComm c;
   std::vector<TcpSock*> socks;
   for (int i = 0; i < 4; i++) {
@@ -49,7 +52,6 @@ int main() {
   srand(7);
   uint64_t posted = 0, completed = 0;
   std::vector<uint32_t> outstanding;  // seqs in flight
-  const uint64_t kTarget = 30'000'000;
   uint64_t spins_since_progress = 0;
   while (completed < kTarget) {
     // post while depth < 12
@@ -104,7 +106,7 @@ int main() {
         SendRequest* r = &c.reqs[seq % NCCL_NET_MAX_REQUESTS];
         fprintf(stderr,
                 "  seq=%u total=%u chunk=%u cur=%u/g%u sent=%u hdrsent=%d\n",
-                seq, r->total, r->chunk, cur_off(r->cursor.load()),
+                seq, r->total.load(), r->chunk.load(), cur_off(r->cursor.load()),
                 cur_gen(r->cursor.load()), r->sent.load(),
                 (int)r->hdr_sent.load());
       }
