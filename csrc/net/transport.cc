@@ -57,37 +57,26 @@ void IoThread::stop() {
 
 void IoThread::add_sock(TcpSock* s) {
   std::lock_guard<std::mutex> lk(task_mu_);
-  tasks_.push_back({Task::ADD, s, nullptr, nullptr});
+  tasks_.push_back({Task::ADD, s, nullptr});
   uint64_t one = 1;
   (void)!write(evfd_, &one, sizeof(one));
 }
 
 void IoThread::remove_sock_sync(TcpSock* s) {
-  std::mutex mu;
-  std::condition_variable cv;
-  bool done = false;
+  auto tok = std::make_shared<SyncToken>();
   {
     std::lock_guard<std::mutex> lk(task_mu_);
-    tasks_.push_back({Task::REMOVE, s, &cv, &done});
+    tasks_.push_back({Task::REMOVE, s, tok});
     uint64_t one = 1;
     (void)!write(evfd_, &one, sizeof(one));
   }
-  std::unique_lock<std::mutex> lk(mu);
-  // The task's cv/flag are signalled under task_mu_; use a simple poll wait
-  // to keep the handshake one-sided.
-  while (true) {
-    {
-      std::lock_guard<std::mutex> l2(task_mu_);
-      if (done) break;
-    }
-    cv.wait_for(lk, std::chrono::milliseconds(1));
-  }
+  tok->wait();
 }
 
 void IoThread::kick(TcpSock* s) {
   s->dbg_kicks.fetch_add(1, std::memory_order_relaxed);
   std::lock_guard<std::mutex> lk(task_mu_);
-  tasks_.push_back({Task::KICK, s, nullptr, nullptr});
+  tasks_.push_back({Task::KICK, s, nullptr});
   uint64_t one = 1;
   (void)!write(evfd_, &one, sizeof(one));
 }
@@ -120,11 +109,7 @@ void IoThread::handle_tasks() {
                      socks_.end());
         if (s->scomm) s->scomm->live_socks.fetch_sub(1);
         if (s->rcomm) s->rcomm->live_socks.fetch_sub(1);
-        {
-          std::lock_guard<std::mutex> lk(task_mu_);
-          *t.flag = true;
-        }
-        t.cv->notify_all();
+        t.tok->signal();
         break;
       }
       case Task::KICK:

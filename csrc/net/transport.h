@@ -161,6 +161,28 @@ struct RecvRequest {
   }
 };
 
+// Completion token for synchronous engine-task handshakes (socket
+// removal).  shared_ptr ownership: the waiter and the IO thread each hold
+// a reference, so neither side can destroy the cv while the other is
+// still inside wait()/notify_all() (a stack-allocated cv here was a
+// TSan-caught destroy-during-broadcast race).
+struct SyncToken {
+  std::mutex mu;
+  std::condition_variable cv;
+  bool done = false;
+  void signal() {
+    {
+      std::lock_guard<std::mutex> lk(mu);
+      done = true;
+    }
+    cv.notify_all();
+  }
+  void wait() {
+    std::unique_lock<std::mutex> lk(mu);
+    cv.wait(lk, [this] { return done; });
+  }
+};
+
 // -------------------------------------------------------------- socket ----
 
 struct TcpSock {
@@ -351,8 +373,7 @@ class IoThread : public IIoThread {
   struct Task {
     enum { ADD, REMOVE, KICK } kind;
     TcpSock* s;
-    std::condition_variable* cv;
-    bool* flag;
+    std::shared_ptr<SyncToken> tok;  // REMOVE only
   };
   std::vector<Task> tasks_;
   std::vector<TcpSock*> socks_;  // owned set (IO thread only)

@@ -162,30 +162,20 @@ class UringIoThread : public IIoThread {
     close(evfd_);
   }
   void add_sock(TcpSock* s) override {
-    enqueue({Task::ADD, s, nullptr, nullptr});
+    enqueue({Task::ADD, s, nullptr});
   }
   void remove_sock_sync(TcpSock* s) override {
-    std::mutex mu;
-    std::condition_variable cv;
-    bool done = false;
-    enqueue({Task::REMOVE, s, &cv, &done});
-    std::unique_lock<std::mutex> lk(mu);
-    while (true) {
-      {
-        std::lock_guard<std::mutex> l2(task_mu_);
-        if (done) break;
-      }
-      cv.wait_for(lk, std::chrono::milliseconds(1));
-    }
+    auto tok = std::make_shared<SyncToken>();
+    enqueue({Task::REMOVE, s, tok});
+    tok->wait();
   }
-  void kick(TcpSock* s) override { enqueue({Task::KICK, s, nullptr, nullptr}); }
+  void kick(TcpSock* s) override { enqueue({Task::KICK, s, nullptr}); }
 
  private:
   struct Task {
     enum { ADD, REMOVE, KICK } kind;
     TcpSock* s;
-    std::condition_variable* cv;
-    bool* flag;
+    std::shared_ptr<SyncToken> tok;  // REMOVE only
   };
 
   void enqueue(Task t) {
@@ -421,11 +411,7 @@ class UringIoThread : public IIoThread {
                        socks_.end());
           if (s->scomm) s->scomm->live_socks.fetch_sub(1);
           if (s->rcomm) s->rcomm->live_socks.fetch_sub(1);
-          {
-            std::lock_guard<std::mutex> lk(task_mu_);
-            *t.flag = true;
-          }
-          t.cv->notify_all();
+          t.tok->signal();
           break;
         }
         case Task::KICK:
