@@ -32,3 +32,14 @@ def test_native_tsan():
     assert "claim stress ok" in res.stdout
     assert res.stdout.count("loopback soak ok") == 2
     assert "ThreadSanitizer" not in res.stdout + res.stderr
+
+
+def test_native_asan():
+    res = subprocess.run(
+        ["make", "asan-test"], cwd=REPO / "csrc",
+        capture_output=True, text=True, timeout=900,
+    )
+    assert res.returncode == 0, res.stdout + res.stderr
+    assert res.stdout.count("loopback soak ok") == 2
+    assert "AddressSanitizer" not in res.stdout + res.stderr
+    assert "LeakSanitizer" not in res.stdout + res.stderr
