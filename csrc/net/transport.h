@@ -218,6 +218,7 @@ struct TcpSock {
   struct {
     uint8_t op = 0;  // 0 none, 1 send(writev), 2 recv
     bool closing = false;  // removal in progress: no resubmission
+    bool eof = false;      // recv: orderly peer shutdown — never re-arm
     // batched send: up to kUrBatch chunks in one ordered WRITEV
     int nchunks = 0;
     ChunkHdr hdrs[8];

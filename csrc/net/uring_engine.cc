@@ -344,8 +344,10 @@ class UringIoThread : public IIoThread {
     }
     if (res == 0) {  // EOF
       if (!s->rx.in_payload && s->rx.hdr_got == 0 &&
-          !recv_socket_incomplete(c))
-        return;  // orderly shutdown: just stop re-arming
+          !recv_socket_incomplete(c)) {
+        s->ur.eof = true;
+        return;  // orderly shutdown: never re-arm
+      }
       c->error.store(ECONNRESET);
       BNET_WARN("bnet(uring) recv eof mid-protocol");
       return;
@@ -479,10 +481,20 @@ class UringIoThread : public IIoThread {
       drain_cqes();
       handle_tasks();
       if (spin || had) {
-        // jobs/watermarks may have advanced: retry idle senders
-        for (TcpSock* s : socks_)
-          if (!s->is_recv && !s->ur.op && s->ur.nchunks == 0)
+        // jobs/watermarks may have advanced: retry any socket without an
+        // op in flight.  Senders with a prepared batch (nchunks > 0) are
+        // included — a ring-full get_sqe() at submit time would otherwise
+        // strand the batch until an external kick (reachable only with
+        // >255 sockets on one thread, but cheap to close).
+        for (TcpSock* s : socks_) {
+          if (s->ur.op) continue;
+          if (s->is_recv) {
+            if (!s->ur.eof && !s->parked.load(std::memory_order_relaxed))
+              submit_recv(s);
+          } else {
             submit_send(s);
+          }
+        }
         ring_.enter(0);
       }
     }
