@@ -450,3 +450,49 @@ def test_roundtrip_config_matrix():
             "BNET_IMPLEMENT": eng,
         }
         assert _run_sub(_roundtrip_probe, env) == "ok", (ns, iot, mc, eng)
+
+
+def _churn_probe(env, q):
+    """200 establish/transfer/close cycles: stresses engine add/remove
+    (the SyncToken teardown handshake) and listen-socket lifecycle."""
+    for k, v in env.items():
+        os.environ[k] = v
+    import ctypes as C
+
+    from baguanet.plugin import Plugin
+    from tests.test_plugin_loopback import establish
+
+    p = Plugin()
+    for cycle in range(200):
+        lcomm, scomm, rcomm = establish(p)
+        smh = p.reg_mr(scomm, None, 0)
+        rmh = p.reg_mr(rcomm, None, 0)
+        size = (cycle * 7919) % 100_000
+        payload = bytes((cycle + j) % 256 for j in range(size))
+        sbuf = C.create_string_buffer(payload, max(size, 1))
+        rbuf = C.create_string_buffer(size + 1)
+        rreq = p.irecv(rcomm, rbuf, size, rmh)
+        sreq = p.isend(scomm, sbuf, size, smh)
+        assert rreq is not None and sreq is not None
+        import time as _t
+
+        t0 = _t.monotonic()
+        sdone = rdone = False
+        while not (sdone and rdone):
+            if not sdone:
+                sdone, _ = p.test(sreq)
+            if not rdone:
+                rdone, got = p.test(rreq)
+            assert _t.monotonic() - t0 < 30, f"cycle {cycle} stalled"
+        assert rbuf.raw[:size] == payload
+        p.close_send(scomm)
+        p.close_recv(rcomm)
+        p.close_listen(lcomm)
+    q.put("ok")
+
+
+def test_comm_churn():
+    for eng in ("EPOLL", "URING"):
+        env = {"NCCL_SOCKET_IFNAME": "lo", "BNET_NSTREAMS": "4",
+               "BNET_IMPLEMENT": eng}
+        assert _run_sub(_churn_probe, env) == "ok", eng
