@@ -95,3 +95,44 @@ def test_stress_random_traffic(plugin):
         plugin.close_send(sc)
         plugin.close_recv(rc)
         plugin.close_listen(lc)
+
+
+def test_many_parallel_comms(plugin):
+    """32 comm pairs alive at once, one verified message each, closed in
+    an interleaved order — stresses conn_id grouping in accept and the
+    engines' shared-socket bookkeeping."""
+    import ctypes as C
+    import time
+
+    pairs = [establish(plugin) for _ in range(32)]
+    msgs = []
+    for k, (lcomm, scomm, rcomm) in enumerate(pairs):
+        smh = plugin.reg_mr(scomm, None, 0)
+        rmh = plugin.reg_mr(rcomm, None, 0)
+        size = 1000 * (k + 1)
+        payload = bytes((k + j) % 256 for j in range(size))
+        sbuf = C.create_string_buffer(payload, size)
+        rbuf = C.create_string_buffer(size + 1)
+        rreq = plugin.irecv(rcomm, rbuf, size, rmh)
+        sreq = plugin.isend(scomm, sbuf, size, smh)
+        assert rreq is not None and sreq is not None
+        msgs.append([sreq, rreq, False, False, size, payload, sbuf, rbuf])
+    t0 = time.monotonic()
+    while any(not (m[2] and m[3]) for m in msgs):
+        for m in msgs:
+            if not m[2]:
+                m[2], _ = plugin.test(m[0])
+            if not m[3]:
+                done, got = plugin.test(m[1])
+                if done:
+                    assert got == m[4]
+                    m[3] = True
+        assert time.monotonic() - t0 < 30
+    for m in msgs:
+        assert m[7].raw[:m[4]] == m[5]
+    # interleaved close order (even pairs first, then odd)
+    for k in list(range(0, 32, 2)) + list(range(1, 32, 2)):
+        lcomm, scomm, rcomm = pairs[k]
+        plugin.close_send(scomm)
+        plugin.close_recv(rcomm)
+        plugin.close_listen(lcomm)
