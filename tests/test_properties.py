@@ -3,6 +3,7 @@ partitioning, and wire-protocol roundtrips over randomized inputs with
 shrinking.  Complements the fixed-seed stress tests."""
 
 import ctypes as C
+import os
 
 import torch
 from hypothesis import given, settings, strategies as st
@@ -10,8 +11,12 @@ from hypothesis import given, settings, strategies as st
 from baguanet.optim import FusedSGD
 from baguanet.parallel import BucketedDDP
 
+# BNET_HYP_EXAMPLES scales fuzz depth (CI default is light; deep one-off
+# campaigns set it to hundreds)
+N = int(os.environ.get("BNET_HYP_EXAMPLES", "25"))
 
-@settings(max_examples=25, deadline=None)
+
+@settings(max_examples=N, deadline=None)
 @given(
     lr=st.floats(1e-4, 1.0),
     momentum=st.sampled_from([0.0, 0.5, 0.9]),
@@ -39,7 +44,7 @@ def test_fused_sgd_matches_torch(lr, momentum, wd, nesterov, steps, shape):
     assert torch.allclose(p_our.double(), p_ref, rtol=1e-4, atol=1e-5)
 
 
-@settings(max_examples=25, deadline=None)
+@settings(max_examples=N, deadline=None)
 @given(
     layer_sizes=st.lists(st.integers(1, 300), min_size=1, max_size=12),
     cap_kb=st.sampled_from([1, 4, 64, 1024]),
@@ -87,7 +92,7 @@ def test_bucket_partition_invariants(layer_sizes, cap_kb):
         assert p.grad.shape == p.shape
 
 
-@settings(max_examples=10, deadline=None)
+@settings(max_examples=max(N // 2, 5), deadline=None)
 @given(
     sizes=st.lists(st.integers(0, 2 * 1024 * 1024), min_size=1, max_size=8),
     drains=st.lists(st.booleans(), min_size=8, max_size=8),
