@@ -65,32 +65,33 @@ def _worker(rank, world, port, q):
         dist.destroy_process_group()
 
 
-def _reference():
-    """Single-process equivalent: average of the two ranks' gradients."""
+def _reference(world=2):
+    """Single-process equivalent: average of the ranks' gradients."""
     model = _model(seed=7)
     for step in range(3):
         for p in model.parameters():
             p.grad = None
-        total = None
         gs = []
-        for rank in range(2):
+        for rank in range(world):
             x, y = _data(100 + rank)
             loss = nn.functional.cross_entropy(model(x), y)
             g = torch.autograd.grad(loss, list(model.parameters()))
             gs.append(g)
-        avg = [(a + b) / 2 for a, b in zip(gs[0], gs[1])]
+        avg = [sum(t) / world for t in zip(*gs)]
         with torch.no_grad():
             for p, g in zip(model.parameters(), avg):
                 p -= 0.1 * g
     return avg, [p.detach().clone() for p in model.parameters()]
 
 
-def test_bucket_ddp_matches_reference():
+@pytest.mark.parametrize("world", [2, 4])
+def test_bucket_ddp_matches_reference(world):
     port = _free_port()
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
     procs = [
-        ctx.Process(target=_worker, args=(r, 2, port, q)) for r in range(2)
+        ctx.Process(target=_worker, args=(r, world, port, q))
+        for r in range(world)
     ]
     for p in procs:
         p.start()
@@ -99,7 +100,7 @@ def test_bucket_ddp_matches_reference():
         p.join(60)
         assert p.exitcode == 0
     assert tag == "ok"
-    ref_grads, ref_params = _reference()
+    ref_grads, ref_params = _reference(world)
     for g, rg in zip(grads, ref_grads):
         torch.testing.assert_close(torch.from_numpy(g), rg, rtol=1e-5,
                                    atol=1e-6)
