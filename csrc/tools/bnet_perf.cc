@@ -13,6 +13,7 @@
 #include <sys/wait.h>
 #include <unistd.h>
 
+#include <algorithm>
 #include <chrono>
 #include <cstdio>
 #include <cstdlib>
@@ -140,11 +141,83 @@ void run_sender(void* handle, const std::vector<size_t>& sizes,
   CHECK(net->closeSend(scomm));
 }
 
+// ---- latency (ping-pong RTT) ---------------------------------------------
+// Bidirectional: each process owns one send comm and one recv comm.  The
+// echo side mirrors every message; the measuring side records RTTs and
+// prints percentiles.  One message in flight — measures per-message
+// latency, not throughput.
+
+void setup_duplex(void* my_handle, void* my_lcomm, int peer_handle_fd,
+                  void** scomm, void** rcomm) {
+  char peer[NCCL_NET_HANDLE_MAXSIZE];
+  size_t got = 0;
+  while (got < sizeof(peer)) {
+    ssize_t n = read(peer_handle_fd, peer + got, sizeof(peer) - got);
+    if (n <= 0) exit(1);
+    got += (size_t)n;
+  }
+  (void)my_handle;
+  *scomm = nullptr;
+  *rcomm = nullptr;
+  while (!*scomm || !*rcomm) {
+    if (!*scomm) CHECK(net->connect(0, peer, scomm));
+    if (!*rcomm) CHECK(net->accept(my_lcomm, rcomm));
+  }
+}
+
+void run_latency(void* scomm, void* rcomm, const std::vector<size_t>& sizes,
+                 int iters, bool measure) {
+  void* smh = nullptr;
+  void* rmh = nullptr;
+  CHECK(net->regMr(scomm, nullptr, 0, NCCL_PTR_HOST, &smh));
+  CHECK(net->regMr(rcomm, nullptr, 0, NCCL_PTR_HOST, &rmh));
+  for (size_t size : sizes) {
+    std::vector<char> sbuf(std::max(size, 1ul), 0x5a);
+    std::vector<char> rbuf(std::max(size, 1ul));
+    std::vector<double> rtts;
+    rtts.reserve(iters);
+    int warmup = iters / 10 + 8;
+    for (int it = 0; it < iters + warmup; it++) {
+      void* rreq = nullptr;
+      void* data = rbuf.data();
+      int sz = (int)size;
+      int tag = 0;
+      while (!rreq) CHECK(net->irecv(rcomm, 1, &data, &sz, &tag, &rmh, &rreq));
+      double t0 = now_s();
+      if (measure) {  // ping → wait echo
+        void* sreq = nullptr;
+        while (!sreq) CHECK(net->isend(scomm, sbuf.data(), (int)size, 0,
+                                       smh, &sreq));
+        wait_req(sreq);
+        wait_req(rreq);
+        if (it >= warmup) rtts.push_back(now_s() - t0);
+      } else {  // echo: wait ping → reply
+        wait_req(rreq);
+        void* sreq = nullptr;
+        while (!sreq) CHECK(net->isend(scomm, sbuf.data(), (int)size, 0,
+                                       smh, &sreq));
+        wait_req(sreq);
+      }
+    }
+    if (measure) {
+      std::sort(rtts.begin(), rtts.end());
+      auto pct = [&](double p) {
+        return rtts[std::min(rtts.size() - 1,
+                             (size_t)(p * rtts.size()))] * 1e6;
+      };
+      printf("%10zu B x %6d iters: RTT p50 %7.1f us  p90 %7.1f us  "
+             "p99 %7.1f us  min %7.1f us\n",
+             size, iters, pct(0.50), pct(0.90), pct(0.99), rtts.front() * 1e6);
+      fflush(stdout);
+    }
+  }
+}
+
 }  // namespace
 
 int main(int argc, char** argv) {
   if (argc < 2) {
-    fprintf(stderr, "usage: %s loop|recv|send [sizes...]\n", argv[0]);
+    fprintf(stderr, "usage: %s loop|lat [sizes...]\n", argv[0]);
     return 2;
   }
   CHECK(net->init(nullptr));
@@ -170,6 +243,56 @@ int main(int argc, char** argv) {
     CHECK(net->closeListen(lcomm));
     return WIFEXITED(st) ? WEXITSTATUS(st) : 1;
   }
-  fprintf(stderr, "only 'loop' mode is wired up in this build\n");
+  if (!strcmp(argv[1], "lat")) {
+    auto sizes = parse_sizes(argc, argv, 2);
+    if (argc <= 2) sizes = {8, 4096, 65536};
+    int iters = 2000;
+    if (const char* e = getenv("BNET_PERF_ITERS")) iters = atoi(e);
+    char handle1[NCCL_NET_HANDLE_MAXSIZE] = {};
+    void* lcomm1 = nullptr;
+    CHECK(net->listen(0, handle1, &lcomm1));
+    int c2p[2];  // child sends its listen handle to the parent
+    if (pipe(c2p)) return 1;
+    pid_t pid = fork();
+    if (pid == 0) {  // echo side: own listener, handle over the pipe
+      char handle2[NCCL_NET_HANDLE_MAXSIZE] = {};
+      void* lcomm2 = nullptr;
+      CHECK(net->listen(0, handle2, &lcomm2));
+      (void)!write(c2p[1], handle1, sizeof(handle1));  // unused, keeps sym
+      (void)!write(c2p[1], handle2, sizeof(handle2));
+      void* scomm = nullptr;
+      void* rcomm = nullptr;
+      // child connects to handle1, accepts on lcomm2
+      scomm = nullptr;
+      while (!scomm || !rcomm) {
+        if (!scomm) CHECK(net->connect(0, handle1, &scomm));
+        if (!rcomm) CHECK(net->accept(lcomm2, &rcomm));
+      }
+      run_latency(scomm, rcomm, sizes, iters, /*measure=*/false);
+      CHECK(net->closeSend(scomm));
+      CHECK(net->closeRecv(rcomm));
+      _exit(0);
+    }
+    char dummy[NCCL_NET_HANDLE_MAXSIZE];
+    void* scomm = nullptr;
+    void* rcomm = nullptr;
+    {
+      size_t got = 0;
+      while (got < sizeof(dummy)) {
+        ssize_t n = read(c2p[0], dummy + got, sizeof(dummy) - got);
+        if (n <= 0) return 1;
+        got += (size_t)n;
+      }
+    }
+    setup_duplex(handle1, lcomm1, c2p[0], &scomm, &rcomm);
+    run_latency(scomm, rcomm, sizes, iters, /*measure=*/true);
+    CHECK(net->closeSend(scomm));
+    CHECK(net->closeRecv(rcomm));
+    int st = 0;
+    waitpid(pid, &st, 0);
+    CHECK(net->closeListen(lcomm1));
+    return WIFEXITED(st) ? WEXITSTATUS(st) : 1;
+  }
+  fprintf(stderr, "only 'loop' and 'lat' modes are wired up in this build\n");
   return 2;
 }
