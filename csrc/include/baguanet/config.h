@@ -40,6 +40,10 @@ struct Config {
   int stage_kernel = 0;
   // Listen backlog (reference: 16384, nthread:101).
   int backlog = 16384;
+  // Busy-spin window after the last observed traffic before an IO
+  // thread blocks, in microseconds.  Spinning keeps latency-critical
+  // ping-pong patterns off the scheduler wake path.
+  uint32_t spin_us = 200;
   // IO engine: "EPOLL" (default) or "URING" (io_uring; falls back to
   // epoll when unavailable).  The reference's BAGUA_NET_IMPLEMENT
   // BASIC/TOKIO selector, rebuilt as readiness- vs completion-based IO.

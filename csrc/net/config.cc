@@ -39,6 +39,7 @@ const Config& Config::get() {
     if (c.stage_chunk < 65536) c.stage_chunk = 65536;
     c.stage_kernel = (int)env_long("BNET_STAGE_KERNEL", 0);
     c.backlog = (int)env_long("BNET_BACKLOG", c.backlog);
+    c.spin_us = (uint32_t)env_long("BNET_SPIN_US", c.spin_us);
     c.implement = env_str("BNET_IMPLEMENT", "EPOLL");
     c.metrics_file = env_str("BNET_METRICS_FILE", "");
     c.trace_file = env_str("BNET_TRACE_FILE", "");

@@ -146,7 +146,8 @@ void IoThread::run() {
       }
     }
     uint64_t now = now_ns();
-    bool spin = staging_busy || (now - last_active_ns < 200'000);
+    bool spin = staging_busy ||
+                  (now - last_active_ns < Config::get().spin_us * 1000ull);
     int n = epoll_wait(epfd_, evs, 64, spin ? 0 : 100);
     if (n < 0 && errno != EINTR) break;
     if (n > 0) last_active_ns = now;

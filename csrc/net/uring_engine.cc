@@ -472,7 +472,8 @@ class UringIoThread : public IIoThread {
         }
       }
       uint64_t now = now_ns();
-      bool spin = staging_busy || (now - last_active_ns < 200'000);
+      bool spin = staging_busy ||
+                  (now - last_active_ns < Config::get().spin_us * 1000ull);
       if (!evfd_armed_) arm_eventfd();
       int rc = ring_.enter(spin ? 0 : 1);
       (void)rc;

@@ -39,6 +39,7 @@ def test_env_knobs_respected():
                 "BNET_MIN_CHUNKSIZE": "65536",
                 "BNET_MAX_CHUNKSIZE": "2097152",
                 "BNET_IO_THREADS": "3",
+                "BNET_SPIN_US": "77",
                 "BNET_STAGE_KERNEL": "1",
             },
         )
@@ -47,6 +48,7 @@ def test_env_knobs_respected():
     assert cfg["min_chunk"] == 65536
     assert cfg["max_chunk"] == 2097152
     assert cfg["io_threads"] == 3
+    assert cfg["spin_us"] == 77
     assert cfg["stage_kernel"] == 1
 
 
