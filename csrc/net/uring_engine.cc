@@ -257,6 +257,37 @@ class UringIoThread : public IIoThread {
       finish_batch(s);
       return;
     }
+    // Inline-first: try one direct nonblocking writev before arming a
+    // ring op.  On an idle ring, CQE delivery costs ~20 us per wakeup
+    // (poll-arm -> task_work), which dominated single-message latency
+    // (measured 70 us RTT vs 17 us on epoll); writing synchronously from
+    // the kicked thread puts the bytes on the wire immediately, and the
+    // ring carries only the remainder when the socket buffer fills.
+    // Mirrors the receive side, where drain_recv already reads greedily
+    // and the RECV op only waits for new data.
+    ssize_t n = writev(s->fd, s->ur.iov, iovn);
+    if (n > 0) {
+      s->ur.done += (uint32_t)n;
+      if (s->ur.done == s->ur.batch_bytes) {
+        finish_batch(s);
+        submit_send(s);  // claim + send the next batch inline too
+        return;
+      }
+      // partial: rebuild the iovec for the unwritten tail
+      submit_send_tail(s);
+      return;
+    }
+    if (n < 0 && errno != EAGAIN && errno != EWOULDBLOCK &&
+        errno != EINTR) {
+      s->scomm->error.store(errno);
+      BNET_WARN("bnet(uring) inline send error: %s", strerror(errno));
+      return;
+    }
+    arm_writev(s, iovn);
+  }
+
+  // arm the WRITEV SQE for the current iov (already built)
+  void arm_writev(TcpSock* s, int iovn) {
     io_uring_sqe* sqe = ring_.get_sqe();
     if (!sqe) return;  // ring full; retried after next drain
     sqe->opcode = IORING_OP_WRITEV;
@@ -266,6 +297,32 @@ class UringIoThread : public IIoThread {
     sqe->user_data = (uint64_t)(uintptr_t)s;
     ring_.advance_tail();
     s->ur.op = 1;
+  }
+
+  // rebuild the iov for the unwritten tail of the batch and arm it
+  void submit_send_tail(TcpSock* s) {
+    int iovn = 0;
+    uint32_t skip = s->ur.done;
+    for (int i = 0; i < s->ur.nchunks; i++) {
+      uint32_t span = 16 + s->ur.hdrs[i].len;
+      if (skip >= span) {
+        skip -= span;
+        continue;
+      }
+      uint32_t o = skip;
+      skip = 0;
+      if (o < 16)
+        s->ur.iov[iovn++] = {(char*)&s->ur.hdrs[i] + o, 16 - o};
+      uint32_t pay_off = o > 16 ? o - 16 : 0;
+      if (s->ur.hdrs[i].len > pay_off)
+        s->ur.iov[iovn++] = {(void*)(s->ur.payloads[i] + pay_off),
+                             s->ur.hdrs[i].len - pay_off};
+    }
+    if (iovn == 0) {
+      finish_batch(s);
+      return;
+    }
+    arm_writev(s, iovn);
   }
 
   void finish_batch(TcpSock* s) {
