@@ -12,11 +12,14 @@ from baguanet.optim import FusedSGD
 from baguanet.parallel import BucketedDDP
 
 # BNET_HYP_EXAMPLES scales fuzz depth (CI default is light; deep one-off
-# campaigns set it to hundreds)
+# campaigns set it to hundreds).  CI runs derandomized so the driver's
+# suite can't redden on a never-before-seen example; exploratory
+# campaigns set BNET_HYP_DERANDOMIZE=0.
 N = int(os.environ.get("BNET_HYP_EXAMPLES", "25"))
+DERAND = os.environ.get("BNET_HYP_DERANDOMIZE", "1") == "1"
 
 
-@settings(max_examples=N, deadline=None)
+@settings(max_examples=N, deadline=None, derandomize=DERAND)
 @given(
     lr=st.floats(1e-4, 1.0),
     momentum=st.sampled_from([0.0, 0.5, 0.9]),
@@ -44,7 +47,7 @@ def test_fused_sgd_matches_torch(lr, momentum, wd, nesterov, steps, shape):
     assert torch.allclose(p_our.double(), p_ref, rtol=1e-4, atol=1e-5)
 
 
-@settings(max_examples=N, deadline=None)
+@settings(max_examples=N, deadline=None, derandomize=DERAND)
 @given(
     layer_sizes=st.lists(st.integers(1, 300), min_size=1, max_size=12),
     cap_kb=st.sampled_from([1, 4, 64, 1024]),
@@ -92,7 +95,7 @@ def test_bucket_partition_invariants(layer_sizes, cap_kb):
         assert p.grad.shape == p.shape
 
 
-@settings(max_examples=max(N // 2, 5), deadline=None)
+@settings(max_examples=max(N // 2, 5), deadline=None, derandomize=DERAND)
 @given(
     sizes=st.lists(st.integers(0, 2 * 1024 * 1024), min_size=1, max_size=8),
     drains=st.lists(st.booleans(), min_size=8, max_size=8),
