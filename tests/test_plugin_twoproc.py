@@ -28,8 +28,10 @@ def _set_env():
     os.environ["BNET_NSTREAMS"] = "3"
 
 
-def _receiver(conn, result):
+def _receiver(conn, result, engine=None):
     _set_env()
+    if engine:
+        os.environ["BNET_IMPLEMENT"] = engine
     import ctypes as C
 
     from baguanet.plugin import Plugin
@@ -56,8 +58,10 @@ def _receiver(conn, result):
     p.close_listen(lcomm)
 
 
-def _sender(conn, result):
+def _sender(conn, result, engine=None):
     _set_env()
+    if engine:
+        os.environ["BNET_IMPLEMENT"] = engine
     import ctypes as C
 
     from baguanet.plugin import Plugin
@@ -82,12 +86,12 @@ def _sender(conn, result):
     p.close_send(scomm)
 
 
-def _run_pair():
+def _run_pair(send_engine=None, recv_engine=None):
     ctx = mp.get_context("spawn")
     a, b = ctx.Pipe()
     res = ctx.Queue()
-    pr = ctx.Process(target=_receiver, args=(a, res))
-    ps = ctx.Process(target=_sender, args=(b, res))
+    pr = ctx.Process(target=_receiver, args=(a, res, recv_engine))
+    ps = ctx.Process(target=_sender, args=(b, res, send_engine))
     pr.start()
     ps.start()
     outs = [res.get(timeout=120), res.get(timeout=120)]
@@ -122,3 +126,12 @@ def test_two_process_transfer_real_nic():
         _run_pair()
     finally:
         del os.environ["BNET_TEST_IFNAME"]
+
+
+def test_cross_engine_interop():
+    """The wire format is engine-independent: epoll sender <-> io_uring
+    receiver and vice versa must interoperate byte-perfectly (the
+    reference's BASIC and TOKIO backends could NOT interoperate — their
+    length frames differed)."""
+    _run_pair(send_engine="EPOLL", recv_engine="URING")
+    _run_pair(send_engine="URING", recv_engine="EPOLL")
