@@ -40,6 +40,9 @@ struct Config {
   int stage_kernel = 0;
   // Listen backlog (reference: 16384, nthread:101).
   int backlog = 16384;
+  // Accepted connections must complete their WireHello within this
+  // window or be reaped (dead client protection on long-lived listeners).
+  uint32_t hello_timeout_ms = 30000;
   // Busy-spin window after the last observed traffic before an IO
   // thread blocks, in microseconds.  Spinning keeps latency-critical
   // ping-pong patterns off the scheduler wake path.

@@ -40,6 +40,8 @@ const Config& Config::get() {
     c.stage_kernel = (int)env_long("BNET_STAGE_KERNEL", 0);
     c.backlog = (int)env_long("BNET_BACKLOG", c.backlog);
     c.spin_us = (uint32_t)env_long("BNET_SPIN_US", c.spin_us);
+    c.hello_timeout_ms = (uint32_t)env_long("BNET_HELLO_TIMEOUT_MS",
+                                            c.hello_timeout_ms);
     c.implement = env_str("BNET_IMPLEMENT", "EPOLL");
     c.metrics_file = env_str("BNET_METRICS_FILE", "");
     c.trace_file = env_str("BNET_TRACE_FILE", "");

@@ -261,11 +261,12 @@ ncclResult_t Net::accept(void* listen_comm, void** recv_comm) {
   auto* l = (ListenComm*)listen_comm;
 
   // 1. accept any pending connections
+  uint64_t now = now_ns();
   while (true) {
     int fd = accept4(l->fd, nullptr, nullptr, SOCK_NONBLOCK | SOCK_CLOEXEC);
     if (fd < 0) break;
     tune_socket2(fd);
-    l->half.push_back({fd, {}, 0});
+    l->half.push_back({fd, {}, 0, now});
   }
 
   // 2. progress hello reads
@@ -298,6 +299,15 @@ ncclResult_t Net::accept(void* listen_comm, void** recv_comm) {
       it = l->half.erase(it);
     } else if (n == 0 || (n < 0 && errno != EAGAIN && errno != EWOULDBLOCK &&
                           errno != EINTR)) {
+      close(it->fd);
+      it = l->half.erase(it);
+    } else if (now - it->t0_ns >
+               Config::get().hello_timeout_ms * 1'000'000ull) {
+      // a peer that never finishes its hello must not leak an fd for the
+      // listener's lifetime (dead client / port scanner)
+      BNET_WARN("accept: reaping stalled half-connection (%u/%zu hello "
+                "bytes after %u ms)", it->got, sizeof(WireHello),
+                Config::get().hello_timeout_ms);
       close(it->fd);
       it = l->half.erase(it);
     } else {

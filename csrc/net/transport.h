@@ -309,6 +309,7 @@ struct ListenComm {
     int fd;
     WireHello hello;
     uint32_t got = 0;
+    uint64_t t0_ns = 0;  // accept time — stalled hellos are reaped
   };
   std::vector<HalfConn> half;
   // complete hellos grouped by conn_id

@@ -498,3 +498,58 @@ def test_comm_churn():
         env = {"NCCL_SOCKET_IFNAME": "lo", "BNET_NSTREAMS": "4",
                "BNET_IMPLEMENT": eng}
         assert _run_sub(_churn_probe, env) == "ok", eng
+
+
+def _stalled_hello_reaped(env, q):
+    """A connection that never completes its WireHello is reaped after
+    BNET_HELLO_TIMEOUT_MS and a legitimate connect still succeeds."""
+    for k, v in env.items():
+        os.environ[k] = v
+    import ctypes as C
+    import socket
+    import struct
+    import time
+
+    from baguanet.plugin import Plugin
+
+    p = Plugin()
+    handle, lcomm = p.listen(0)
+    magic, family, port_be = struct.unpack_from("<IHH", bytes(handle))
+    port = socket.ntohs(port_be)
+    # half-open: send 3 bytes of hello, then go silent (socket stays open)
+    stalled = socket.create_connection(("127.0.0.1", port), timeout=5)
+    stalled.sendall(b"\xe7\xa4")
+    t0 = time.monotonic()
+    while time.monotonic() - t0 < 1.0:  # > BNET_HELLO_TIMEOUT_MS=300
+        assert p.accept(lcomm) is None
+        time.sleep(0.02)
+    # the listener must have closed its side by now
+    stalled.settimeout(2)
+    assert stalled.recv(1) == b"", "stalled half-conn not reaped"
+    stalled.close()
+    # a real connection still works
+    scomm = rcomm = None
+    t0 = time.monotonic()
+    while scomm is None or rcomm is None:
+        assert time.monotonic() - t0 < 30
+        if scomm is None:
+            scomm = p.connect(0, handle)
+        if rcomm is None:
+            rcomm = p.accept(lcomm)
+    buf = C.create_string_buffer(b"post-reap", 9)
+    rbuf = C.create_string_buffer(10)
+    mh = p.reg_mr(scomm, None, 0)
+    rreq = p.irecv(rcomm, rbuf, 9, mh)
+    sreq = p.isend(scomm, buf, 9, mh)
+    assert p.wait(sreq, 30) == 9 and p.wait(rreq, 30) == 9
+    assert rbuf.raw[:9] == b"post-reap"
+    p.close_send(scomm)
+    p.close_recv(rcomm)
+    p.close_listen(lcomm)
+    q.put("ok")
+
+
+def test_stalled_hello_reaped():
+    env = {"NCCL_SOCKET_IFNAME": "lo", "BNET_NSTREAMS": "2",
+           "BNET_HELLO_TIMEOUT_MS": "300"}
+    assert _run_sub(_stalled_hello_reaped, env) == "ok"
