@@ -150,7 +150,10 @@ void IoThread::run() {
                   (now - last_active_ns < Config::get().spin_us * 1000ull);
     int n = epoll_wait(epfd_, evs, 64, spin ? 0 : 100);
     if (n < 0 && errno != EINTR) break;
-    if (n > 0) last_active_ns = now;
+    // stamp with a FRESH timestamp: `now` predates a blocking wait, and a
+    // stale stamp would fail to arm the spin window for the burst that
+    // just started
+    if (n > 0) last_active_ns = now_ns();
     for (int i = 0; i < n; i++) {
       if (evs[i].data.ptr == nullptr) {
         uint64_t v;

@@ -535,7 +535,7 @@ class UringIoThread : public IIoThread {
       int rc = ring_.enter(spin ? 0 : 1);
       (void)rc;
       io_uring_cqe* had = ring_.peek();
-      if (had) last_active_ns = now;
+      if (had) last_active_ns = now_ns();  // fresh: `now` predates the wait
       drain_cqes();
       handle_tasks();
       if (spin || had) {
