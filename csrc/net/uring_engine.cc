@@ -9,9 +9,12 @@
 //
 // Semantics are identical to the epoll engine: same chunk claiming
 // (claim_chunk), same header processing (process_recv_header), same
-// completion accounting (finish_rx_chunk), same parking rules.  At most
-// one SQE is in flight per socket; its completion immediately claims and
-// submits the next span, so the kernel always has work queued.
+// completion accounting (finish_rx_chunk), same parking rules.  Sends
+// are inline-first: a claimed batch is written with one direct
+// nonblocking writev and the ring carries only the EAGAIN remainder
+// (at most one WRITEV op in flight per socket — concurrent ops would
+// interleave the byte stream).  Receives mirror it: drain_recv reads
+// greedily and the armed RECV op only waits for new data.
 
 #include <errno.h>
 #include <linux/io_uring.h>
