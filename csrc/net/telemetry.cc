@@ -131,10 +131,11 @@ __attribute__((visibility("default"))) int bnet_config_json(char* buf,
                   "\"io_threads\":%d,\"sockbuf\":%d,\"cuda_ptr\":%d,"
                   "\"stage_pool\":%zu,\"stage_chunk\":%u,"
                   "\"stage_kernel\":%d,\"backlog\":%d,\"spin_us\":%u,"
-                  "\"implement\":\"%s\"}",
+                  "\"hello_timeout_ms\":%u,\"implement\":\"%s\"}",
                   c.nstreams, c.min_chunk, c.max_chunk, c.io_threads,
                   c.sockbuf, (int)c.cuda_ptr, c.stage_pool, c.stage_chunk,
-                  c.stage_kernel, c.backlog, c.spin_us, c.implement.c_str());
+                  c.stage_kernel, c.backlog, c.spin_us, c.hello_timeout_ms,
+                  c.implement.c_str());
 }
 
 // Debug: dump a recv comm's request slots + socket rx state (stall

@@ -40,6 +40,7 @@ def test_env_knobs_respected():
                 "BNET_MAX_CHUNKSIZE": "2097152",
                 "BNET_IO_THREADS": "3",
                 "BNET_SPIN_US": "77",
+                "BNET_HELLO_TIMEOUT_MS": "12345",
                 "BNET_STAGE_KERNEL": "1",
             },
         )
@@ -49,6 +50,7 @@ def test_env_knobs_respected():
     assert cfg["max_chunk"] == 2097152
     assert cfg["io_threads"] == 3
     assert cfg["spin_us"] == 77
+    assert cfg["hello_timeout_ms"] == 12345
     assert cfg["stage_kernel"] == 1
 
 
