@@ -555,3 +555,50 @@ def test_stalled_hello_reaped():
     env = {"NCCL_SOCKET_IFNAME": "lo", "BNET_NSTREAMS": "2",
            "BNET_HELLO_TIMEOUT_MS": "300"}
     assert _run_sub(_stalled_hello_reaped, env) == "ok"
+
+
+def _connect_dead_listener(env, q):
+    """connect() to a listener that has already closed must surface an
+    error (not hang, not crash) and leave the plugin usable."""
+    for k, v in env.items():
+        os.environ[k] = v
+    import ctypes as C
+    import time
+
+    from baguanet.plugin import Plugin
+
+    p = Plugin()
+    handle, lcomm = p.listen(0)
+    handle_bytes = bytes(handle)
+    p.close_listen(lcomm)  # port now dead (refused)
+    dead = (C.c_char * len(handle_bytes)).from_buffer_copy(handle_bytes)
+    t0 = time.monotonic()
+    saw_error = False
+    while time.monotonic() - t0 < 20:
+        try:
+            scomm = p.connect(0, dead)
+        except RuntimeError:
+            saw_error = True  # ncclRemoteError surfaced
+            break
+        assert scomm is None  # must never "succeed"
+    assert saw_error, "connect to dead listener neither errored nor refused"
+    # plugin still usable afterwards
+    from tests.test_plugin_loopback import establish
+
+    l2, s2, r2 = establish(p)
+    buf = C.create_string_buffer(b"alive", 5)
+    rbuf = C.create_string_buffer(6)
+    mh = p.reg_mr(s2, None, 0)
+    rreq = p.irecv(r2, rbuf, 5, mh)
+    sreq = p.isend(s2, buf, 5, mh)
+    assert p.wait(sreq, 30) == 5 and p.wait(rreq, 30) == 5
+    assert rbuf.raw[:5] == b"alive"
+    p.close_send(s2)
+    p.close_recv(r2)
+    p.close_listen(l2)
+    q.put("ok")
+
+
+def test_connect_dead_listener():
+    env = {"NCCL_SOCKET_IFNAME": "lo", "BNET_NSTREAMS": "2"}
+    assert _run_sub(_connect_dead_listener, env) == "ok"
