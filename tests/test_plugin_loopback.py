@@ -154,3 +154,57 @@ def test_zero_then_data(conn):
     payload = os.urandom(123457)
     assert xfer(plugin, scomm, rcomm, payload) == payload
     assert xfer(plugin, scomm, rcomm, b"") == b""
+
+
+def test_one_listener_many_connectors(plugin):
+    """RCCL's real shape: ONE listen comm per rank, many peers connecting
+    to the same handle — interleaved, so accept must group half-finished
+    handshakes by conn_id without mixing streams between connectors."""
+    import time
+
+    handle, lcomm = plugin.listen(0)
+    n = 4
+    scomms = [None] * n
+    rcomms = []
+    t0 = time.monotonic()
+    # drive all connects AND accepts interleaved from one thread
+    while any(s is None for s in scomms) or len(rcomms) < n:
+        for i in range(n):
+            if scomms[i] is None:
+                scomms[i] = plugin.connect(0, handle)
+        r = plugin.accept(lcomm)
+        if r is not None:
+            rcomms.append(r)
+        assert time.monotonic() - t0 < 30
+    # accept-completion order vs connect order is NOT contractual, so
+    # verify pairing-agnostically: post a recv on every rcomm, send a
+    # distinct payload on every scomm, and check the received multiset.
+    import ctypes as C
+
+    payloads = [bytes([i ^ 0x5A]) * (10000 + i) for i in range(n)]
+    smh = [plugin.reg_mr(sc, None, 0) for sc in scomms]
+    rmh = [plugin.reg_mr(rc, None, 0) for rc in rcomms]
+    rbufs, rreqs, sreqs = [], [], []
+    for i, rc in enumerate(rcomms):
+        buf = C.create_string_buffer(20000)
+        req = plugin.irecv(rc, buf, 20000, rmh[i])
+        assert req is not None
+        rbufs.append(buf)
+        rreqs.append(req)
+    for i, sc in enumerate(scomms):
+        sbuf = C.create_string_buffer(payloads[i], len(payloads[i]))
+        req = plugin.isend(sc, sbuf, len(payloads[i]), smh[i])
+        assert req is not None
+        sreqs.append((req, sbuf))
+    got = []
+    for i, req in enumerate(rreqs):
+        sz = plugin.wait(req, 30)
+        got.append(rbufs[i].raw[:sz])
+    for req, _ in sreqs:
+        plugin.wait(req, 30)
+    assert sorted(got) == sorted(payloads)
+    for sc in scomms:
+        plugin.close_send(sc)
+    for rc in rcomms:
+        plugin.close_recv(rc)
+    plugin.close_listen(lcomm)
