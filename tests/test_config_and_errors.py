@@ -602,3 +602,42 @@ def _connect_dead_listener(env, q):
 def test_connect_dead_listener():
     env = {"NCCL_SOCKET_IFNAME": "lo", "BNET_NSTREAMS": "2"}
     assert _run_sub(_connect_dead_listener, env) == "ok"
+
+
+def _oversized_send_errors(env, q):
+    """A peer that sends MORE than the posted recv size (an ncclNet
+    contract violation) must error the recv comm, not overrun the
+    buffer or hang."""
+    for k, v in env.items():
+        os.environ[k] = v
+    import ctypes as C
+    import time
+
+    from baguanet.plugin import Plugin
+    from tests.test_plugin_loopback import establish
+
+    p = Plugin()
+    lcomm, scomm, rcomm = establish(p)
+    mh = p.reg_mr(scomm, None, 0)
+    rbuf = C.create_string_buffer(100 + 1)
+    rreq = p.irecv(rcomm, rbuf, 100, mh)       # receiver expects <= 100
+    sbuf = C.create_string_buffer(b"\xAB" * 5000, 5000)
+    sreq = p.isend(scomm, sbuf, 5000, mh)      # sender violates: 5000
+    assert rreq is not None and sreq is not None
+    t0 = time.monotonic()
+    while True:
+        try:
+            done, _ = p.test(rreq)
+        except RuntimeError:
+            break  # EMSGSIZE surfaced as comm error
+        assert not done, "oversized message must not complete"
+        assert time.monotonic() - t0 < 30, "no error surfaced"
+    p.close_send(scomm)
+    p.close_recv(rcomm)
+    p.close_listen(lcomm)
+    q.put("ok")
+
+
+def test_oversized_send_errors():
+    env = {"NCCL_SOCKET_IFNAME": "lo", "BNET_NSTREAMS": "2"}
+    assert _run_sub(_oversized_send_errors, env) == "ok"
