@@ -1,12 +1,12 @@
 // bnet_perf — native point-to-point benchmark + example client for the
 // ncclNetPlugin_v6 vtable (no python overhead; the true transport ceiling).
 //
-//   bnet_perf recv                      # receiver: prints the handle (hex)
-//   bnet_perf send <handle-hex> [sizes] # sender: connects and streams
-//   bnet_perf loop [sizes]              # both roles, fork, over loopback
+//   bnet_perf loop [sizes]   # fork both roles over loopback: one-way GB/s
+//                            # per size, 16-deep pipelined window
+//   bnet_perf lat  [sizes]   # fork ping-pong: RTT p50/p90/p99/min per size
+//                            # (BNET_PERF_ITERS, default 2000)
 //
-// Reports one-way GB/s per size with a 16-deep pipelined window, data
-// verified by per-message checksums.
+// BNET_PERF_BYTES bounds the bytes per size in loop mode (default 1 GiB).
 
 #include <arpa/inet.h>
 #include <string.h>

@@ -67,3 +67,31 @@ def test_metrics_and_spans(tmp_path):
     # atexit dumps also fired
     assert (tmp_path / "metrics.prom").exists()
     assert (tmp_path / "trace.json").exists()
+
+
+def test_debug_exports_smoke(plugin):
+    """bnet_force_kick / bnet_dump_send_state / bnet_dump_recv_state are
+    the stall-triage API (used by the soak's dump): they must be callable
+    on a live comm without crashing and produce non-empty state."""
+    import ctypes as C
+
+    from tests.test_plugin_loopback import establish
+
+    lcomm, scomm, rcomm = establish(plugin)
+    buf = C.create_string_buffer(8192)
+    n = plugin.lib.bnet_dump_send_state(scomm, buf, 8192)
+    assert n > 0 and b"oldest" in buf.value
+    n = plugin.lib.bnet_dump_recv_state(rcomm, buf, 8192)
+    assert n > 0
+    plugin.lib.bnet_force_kick(scomm)  # send-comm only (see telemetry.cc)
+    # comm still functional after forced kicks
+    mh = plugin.reg_mr(scomm, None, 0)
+    sbuf = C.create_string_buffer(b"kicked", 6)
+    rbuf = C.create_string_buffer(7)
+    rreq = plugin.irecv(rcomm, rbuf, 6, mh)
+    sreq = plugin.isend(scomm, sbuf, 6, mh)
+    assert plugin.wait(sreq, 30) == 6 and plugin.wait(rreq, 30) == 6
+    assert rbuf.raw[:6] == b"kicked"
+    plugin.close_send(scomm)
+    plugin.close_recv(rcomm)
+    plugin.close_listen(lcomm)
