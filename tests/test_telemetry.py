@@ -95,3 +95,33 @@ def test_debug_exports_smoke(plugin):
     plugin.close_send(scomm)
     plugin.close_recv(rcomm)
     plugin.close_listen(lcomm)
+
+
+def test_atexit_file_dumps(tmp_path):
+    """BNET_METRICS_FILE / BNET_TRACE_FILE must be written at process
+    exit (the atexit hook) without any explicit dump call."""
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    mfile = tmp_path / "exit_metrics.prom"
+    tfile = tmp_path / "exit_trace.json"
+    code = (
+        "import os, sys, ctypes as C;"
+        f"sys.path.insert(0, {repo!r});"
+        f"sys.path.insert(0, {os.path.join(repo, 'tests')!r});"
+        "from baguanet.plugin import Plugin;"
+        "from test_plugin_loopback import establish, xfer;"
+        "p = Plugin(); l, s, r = establish(p);"
+        "assert xfer(p, s, r, b'x' * 1234) == b'x' * 1234;"
+        "p.close_send(s); p.close_recv(r); p.close_listen(l)"
+    )
+    env = dict(os.environ, NCCL_SOCKET_IFNAME="lo",
+               BNET_METRICS_FILE=str(mfile), BNET_TRACE_FILE=str(tfile))
+    res = subprocess.run([sys.executable, "-c", code], env=env,
+                         capture_output=True, text=True, timeout=120)
+    assert res.returncode == 0, res.stderr[-1500:]
+    assert "bnet_bytes_sent_total 1234" in mfile.read_text()
+    spans = json.loads(tfile.read_text())
+    events = spans["traceEvents"] if isinstance(spans, dict) else spans
+    assert any(ev.get("name") == "isend" for ev in events)
