@@ -124,4 +124,4 @@ def test_atexit_file_dumps(tmp_path):
     assert "bnet_bytes_sent_total 1234" in mfile.read_text()
     spans = json.loads(tfile.read_text())
     events = spans["traceEvents"] if isinstance(spans, dict) else spans
-    assert any(ev.get("name") == "isend" for ev in events)
+    assert any(str(ev.get("name", "")).startswith("isend") for ev in events)
