@@ -578,6 +578,11 @@ class UringIoThread : public IIoThread {
 bool uring_available() {
   static int avail = -1;
   if (avail < 0) {
+    // test hook: exercise the seccomp-fallback path without seccomp
+    if (const char* v = getenv("BNET_FORCE_NO_URING"); v && *v == '1') {
+      avail = 0;
+      return false;
+    }
     io_uring_params p{};
     int fd = sys_uring_setup(4, &p);
     if (fd >= 0) {

@@ -92,3 +92,36 @@ def test_uring_engine_wire_protocol():
     assert q.get(timeout=180) == "ok"
     proc.join(30)
     assert proc.exitcode == 0
+
+
+def test_uring_unavailable_falls_back(tmp_path):
+    """BNET_IMPLEMENT=URING on a host without io_uring (seccomp) must
+    fall back to the epoll engine and still move data (BNET_FORCE_NO_URING
+    exercises the probe-failure path)."""
+    import json
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    code = (
+        "import os, sys, ctypes as C;"
+        f"sys.path.insert(0, {repo!r});"
+        f"sys.path.insert(0, {os.path.join(repo, 'tests')!r});"
+        "from baguanet.plugin import Plugin;"
+        "from test_plugin_loopback import establish, xfer;"
+        "p = Plugin(); l, s, r = establish(p);"
+        "assert xfer(p, s, r, b'f' * 70000) == b'f' * 70000;"
+        "buf = C.create_string_buffer(1024);"
+        "p.lib.bnet_config_json(buf, 1024);"
+        "print(buf.value.decode());"
+        "p.close_send(s); p.close_recv(r); p.close_listen(l)"
+    )
+    env = dict(os.environ, NCCL_SOCKET_IFNAME="lo",
+               BNET_IMPLEMENT="URING", BNET_FORCE_NO_URING="1",
+               BNET_LOG="info")
+    res = subprocess.run([sys.executable, "-c", code], env=env,
+                         capture_output=True, text=True, timeout=120)
+    assert res.returncode == 0, res.stderr[-1500:]
+    # the engine must have REPORTED epoll (fallback), though config still
+    # says URING was requested
+    assert "engine: EPOLL" in res.stderr or "falling back" in res.stderr
