@@ -641,3 +641,31 @@ def _oversized_send_errors(env, q):
 def test_oversized_send_errors():
     env = {"NCCL_SOCKET_IFNAME": "lo", "BNET_NSTREAMS": "2"}
     assert _run_sub(_oversized_send_errors, env) == "ok"
+
+
+def _ipv6_roundtrip(env, q):
+    """NCCL_SOCKET_FAMILY=10 (AF_INET6): the sockaddr_in6 handle/bind/
+    connect paths must carry data end-to-end over ::1."""
+    for k, v in env.items():
+        os.environ[k] = v
+    from baguanet.plugin import Plugin
+    from tests.test_plugin_loopback import establish, xfer
+
+    p = Plugin()
+    if p.ndev() == 0:
+        q.put("skip")  # no IPv6 loopback in this environment
+        return
+    lcomm, scomm, rcomm = establish(p)
+    payload = b"v6" * 30000
+    assert xfer(p, scomm, rcomm, payload) == payload
+    p.close_send(scomm)
+    p.close_recv(rcomm)
+    p.close_listen(lcomm)
+    q.put("ok")
+
+
+def test_ipv6_family():
+    env = {"NCCL_SOCKET_IFNAME": "lo", "NCCL_SOCKET_FAMILY": "10",
+           "BNET_NSTREAMS": "3", "BNET_MIN_CHUNKSIZE": "8192"}
+    res = _run_sub(_ipv6_roundtrip, env)
+    assert res in ("ok", "skip")
