@@ -60,6 +60,10 @@ def main():
     if args.gpu:
         import torch  # noqa: F811
 
+        if not torch.cuda.is_available():
+            sys.exit("--gpu requires a visible GPU "
+                     "(torch.cuda.is_available() is False)")
+
     rng = random.Random(args.seed)
     p = Plugin()
     lcomm, scomm, rcomm = establish(p)
