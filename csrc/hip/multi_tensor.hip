@@ -84,8 +84,13 @@ size_t multi_copy_launch(const PackDesc* host_descs, int ndesc, void* scratch,
 
   // layout: [ndesc PackDesc][nitems WorkItem]
   uint32_t nitems = 0;
-  for (int i = 0; i < ndesc; i++)
-    nitems += (host_descs[i].bytes + tile - 1) / tile;
+  for (int i = 0; i < ndesc; i++) {
+    uint32_t t = (host_descs[i].bytes + tile - 1) / tile;
+    // WorkItem.tile is uint16: indices 0..65535.  With uint32 byte counts
+    // and tile >= 64 KiB this cannot overflow, but guard anyway.
+    if (t > 65536u) return SIZE_MAX;
+    nitems += t;
+  }
   size_t need = sizeof(PackDesc) * ndesc + sizeof(WorkItem) * nitems;
   if (need > scratch_bytes || ndesc > kMaxDesc) return need;
 
@@ -182,8 +187,11 @@ size_t multi_sgd_launch(const SgdDesc* host_descs, int ndesc, void* scratch,
                         hipStream_t stream) {
   constexpr uint32_t kTileElems = 64 * 1024;  // 256 KiB of fp32 per item
   uint32_t nitems = 0;
-  for (int i = 0; i < ndesc; i++)
-    nitems += (host_descs[i].numel + kTileElems - 1) / kTileElems;
+  for (int i = 0; i < ndesc; i++) {
+    uint32_t t = (host_descs[i].numel + kTileElems - 1) / kTileElems;
+    if (t > 65536u) return SIZE_MAX;  // uint16 tile index guard
+    nitems += t;
+  }
   size_t need = sizeof(SgdDesc) * ndesc + sizeof(WorkItem) * nitems;
   if (need > scratch_bytes || ndesc > kMaxDesc) return need;
 

@@ -43,6 +43,12 @@ struct Config {
   // Accepted connections must complete their WireHello within this
   // window or be reaped (dead client protection on long-lived listeners).
   uint32_t hello_timeout_ms = 30000;
+  // Connector-side in-progress connect() state untouched for this long is
+  // considered abandoned by RCCL and its sockets are reaped (0 = never).
+  uint32_t connect_abandon_ms = 120000;
+  // Process rank for telemetry labeling, from BNET_RANK or RANK (the
+  // launcher env; cf. reference nthread:104-107); -1 = unknown.
+  int rank = -1;
   // Busy-spin window after the last observed traffic before an IO
   // thread blocks, in microseconds.  Spinning keeps latency-critical
   // ping-pong patterns off the scheduler wake path.

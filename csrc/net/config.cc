@@ -1,5 +1,6 @@
 #include "baguanet/config.h"
 
+#include <cstdio>
 #include <cstdlib>
 
 namespace baguanet {
@@ -42,9 +43,27 @@ const Config& Config::get() {
     c.spin_us = (uint32_t)env_long("BNET_SPIN_US", c.spin_us);
     c.hello_timeout_ms = (uint32_t)env_long("BNET_HELLO_TIMEOUT_MS",
                                             c.hello_timeout_ms);
+    c.connect_abandon_ms = (uint32_t)env_long("BNET_CONNECT_ABANDON_MS",
+                                              c.connect_abandon_ms);
     c.implement = env_str("BNET_IMPLEMENT", "EPOLL");
+    c.rank = (int)env_long("BNET_RANK", env_long("RANK", -1));
     c.metrics_file = env_str("BNET_METRICS_FILE", "");
     c.trace_file = env_str("BNET_TRACE_FILE", "");
+    // Rank-template the dump paths so N-rank jobs leave N distinguishable
+    // files instead of overwriting one another: "%r" substitutes the rank;
+    // with no template and a known rank, ".r<rank>" is appended.
+    auto rankify = [&](std::string& path) {
+      if (path.empty()) return;
+      auto pos = path.find("%r");
+      char buf[16];
+      snprintf(buf, sizeof(buf), "%d", c.rank < 0 ? 0 : c.rank);
+      if (pos != std::string::npos)
+        path = path.substr(0, pos) + buf + path.substr(pos + 2);
+      else if (c.rank >= 0)
+        path += std::string(".r") + buf;
+    };
+    rankify(c.metrics_file);
+    rankify(c.trace_file);
     return c;
   }();
   return cfg;

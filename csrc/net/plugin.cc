@@ -8,6 +8,7 @@
 #include "baguanet/log.h"
 #include "baguanet/nccl_abi.h"
 #include "staging.h"
+#include "telemetry.h"
 #include "transport.h"
 
 namespace {
@@ -17,6 +18,7 @@ using baguanet::Net;
 ncclResult_t bnInit(ncclDebugLogger_t logFunction) {
   baguanet::g_logger = logFunction;
   Net::get();  // construct: NIC discovery, staging probe
+  baguanet::Telemetry::get();  // registers the atexit metric/trace dumps
   BNET_INFO("baguanet (MI355X-native multi-stream TCP transport) loaded: "
             "%d device(s), ptrSupport 0x%x",
             Net::get().ndev(), Net::get().ptr_support());

@@ -63,34 +63,41 @@ void Telemetry::span_end(uint32_t slot) {
 void Telemetry::dump_metrics(const char* path) {
   FILE* f = fopen(path, "w");
   if (!f) return;
+  // rank label (reference pushed Prometheus with a `rank` label,
+  // nthread:183-211) — empty when the launcher exported no RANK
+  char rl[32] = "";
+  int rank = Config::get().rank;
+  if (rank >= 0) snprintf(rl, sizeof(rl), "{rank=\"%d\"}", rank);
   fprintf(f, "# baguanet transport metrics (prometheus text format)\n");
-  fprintf(f, "bnet_isend_total %llu\n",
+  fprintf(f, "bnet_isend_total%s %llu\n", rl,
           (unsigned long long)isend_count.load());
-  fprintf(f, "bnet_irecv_total %llu\n",
+  fprintf(f, "bnet_irecv_total%s %llu\n", rl,
           (unsigned long long)irecv_count.load());
-  fprintf(f, "bnet_bytes_sent_total %llu\n",
+  fprintf(f, "bnet_bytes_sent_total%s %llu\n", rl,
           (unsigned long long)bytes_sent.load());
-  fprintf(f, "bnet_bytes_recv_total %llu\n",
+  fprintf(f, "bnet_bytes_recv_total%s %llu\n", rl,
           (unsigned long long)bytes_recv.load());
-  fprintf(f, "bnet_send_comms_total %llu\n",
+  fprintf(f, "bnet_send_comms_total%s %llu\n", rl,
           (unsigned long long)send_comms.load());
-  fprintf(f, "bnet_recv_comms_total %llu\n",
+  fprintf(f, "bnet_recv_comms_total%s %llu\n", rl,
           (unsigned long long)recv_comms.load());
-  fprintf(f, "bnet_staged_d2h_bytes_total %llu\n",
+  fprintf(f, "bnet_staged_d2h_bytes_total%s %llu\n", rl,
           (unsigned long long)staged_d2h_bytes.load());
-  fprintf(f, "bnet_staged_h2d_bytes_total %llu\n",
+  fprintf(f, "bnet_staged_h2d_bytes_total%s %llu\n", rl,
           (unsigned long long)staged_h2d_bytes.load());
+  char rli[32] = "";  // histogram label prefix merging with `le`
+  if (rank >= 0) snprintf(rli, sizeof(rli), "rank=\"%d\",", rank);
   const char* names[2] = {"bnet_isend_nbytes", "bnet_irecv_nbytes"};
   std::atomic<uint64_t>* hists[2] = {isend_hist, irecv_hist};
   for (int h = 0; h < 2; h++) {
     uint64_t cum = 0;
     for (int i = 0; i < 7; i++) {
       cum += hists[h][i].load();
-      fprintf(f, "%s_bucket{le=\"%llu\"} %llu\n", names[h],
+      fprintf(f, "%s_bucket{%sle=\"%llu\"} %llu\n", names[h], rli,
               (unsigned long long)kBounds[i], (unsigned long long)cum);
     }
     cum += hists[h][7].load();
-    fprintf(f, "%s_bucket{le=\"+Inf\"} %llu\n", names[h],
+    fprintf(f, "%s_bucket{%sle=\"+Inf\"} %llu\n", names[h], rli,
             (unsigned long long)cum);
   }
   fclose(f);
@@ -109,9 +116,10 @@ void Telemetry::dump_trace(const char* path) {
     if (!first) fprintf(f, ",\n");
     first = false;
     fprintf(f,
-            "{\"name\":\"%s seq=%u %uB\",\"ph\":\"X\",\"pid\":1,"
+            "{\"name\":\"%s seq=%u %uB\",\"ph\":\"X\",\"pid\":%d,"
             "\"tid\":%llu,\"ts\":%.3f,\"dur\":%.3f}",
             s.kind == 0 ? "isend" : "irecv", s.seq, s.nbytes,
+            Config::get().rank < 0 ? 0 : Config::get().rank,
             (unsigned long long)(s.comm & 0xffff), s.t0 / 1000.0,
             (s.t1 - s.t0) / 1000.0);
   }
@@ -131,11 +139,12 @@ __attribute__((visibility("default"))) int bnet_config_json(char* buf,
                   "\"io_threads\":%d,\"sockbuf\":%d,\"cuda_ptr\":%d,"
                   "\"stage_pool\":%zu,\"stage_chunk\":%u,"
                   "\"stage_kernel\":%d,\"backlog\":%d,\"spin_us\":%u,"
-                  "\"hello_timeout_ms\":%u,\"implement\":\"%s\"}",
+                  "\"hello_timeout_ms\":%u,\"connect_abandon_ms\":%u,"
+                  "\"rank\":%d,\"implement\":\"%s\"}",
                   c.nstreams, c.min_chunk, c.max_chunk, c.io_threads,
                   c.sockbuf, (int)c.cuda_ptr, c.stage_pool, c.stage_chunk,
                   c.stage_kernel, c.backlog, c.spin_us, c.hello_timeout_ms,
-                  c.implement.c_str());
+                  c.connect_abandon_ms, c.rank, c.implement.c_str());
 }
 
 // Debug: dump a recv comm's request slots + socket rx state (stall

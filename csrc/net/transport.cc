@@ -149,7 +149,17 @@ void IoThread::run() {
     bool spin = staging_busy ||
                   (now - last_active_ns < Config::get().spin_us * 1000ull);
     int n = epoll_wait(epfd_, evs, 64, spin ? 0 : 100);
-    if (n < 0 && errno != EINTR) break;
+    if (n < 0 && errno != EINTR) {
+      // A hard epoll failure would silently strand every socket on this
+      // thread; surface it as a comm error on all of them and say so.
+      BNET_WARN("bnet-io%d: epoll_wait failed: %s — marking %zu socket(s) "
+                "errored", idx_, strerror(errno), socks_.size());
+      for (TcpSock* s : socks_) {
+        if (s->scomm) s->scomm->error.store(errno ? errno : EIO);
+        if (s->rcomm) s->rcomm->error.store(errno ? errno : EIO);
+      }
+      break;
+    }
     // stamp with a FRESH timestamp: `now` predates a blocking wait, and a
     // stale stamp would fail to arm the spin window for the burst that
     // just started
@@ -329,7 +339,7 @@ void IoThread::progress_send(TcpSock* s) {
       s->tx.active = true;
       s->tx.req = r;
       s->tx.hdr = {ss_seq(r->state_seq.load(std::memory_order_relaxed)), off,
-                   len, r->total};
+                   len, r->total, r->tag};
       s->tx.payload = r->src + off;
       s->tx.done = 0;
     }
@@ -527,6 +537,15 @@ int process_recv_header(TcpSock* s) {
     BNET_WARN("bnet: message (%u B) exceeds posted buffer (%u B)",
               s->rx.hdr.total, r->capacity);
     c->error.store(EMSGSIZE);
+    return -1;
+  }
+  if (s->rx.hdr.tag != r->tag) {
+    // Matching is seq-ordinal (like NCCL's bundled socket transport); the
+    // echoed tag VERIFIES that assumption — a cross-match is a loud
+    // protocol error instead of silent data corruption.
+    BNET_WARN("bnet: tag mismatch on seq=%u (sent %d, posted %d)", seq,
+              s->rx.hdr.tag, r->tag);
+    c->error.store(EPROTO);
     return -1;
   }
   s->rx.req = r;

@@ -53,7 +53,7 @@ struct StageAlloc;  // staging.h
 // ---------------------------------------------------------------- wire ----
 
 constexpr uint32_t kMagic = 0xBA60A4E7;
-constexpr uint32_t kWireVersion = 1;
+constexpr uint32_t kWireVersion = 2;  // v2: ChunkHdr carries the NCCL tag
 
 struct WireHello {  // connector -> acceptor, once per data socket
   uint32_t magic;
@@ -70,8 +70,11 @@ struct ChunkHdr {  // precedes every payload chunk on a data socket
   uint32_t offset;  // byte offset of this chunk within the message
   uint32_t len;     // chunk payload bytes (0 only for empty messages)
   uint32_t total;   // total message bytes
+  int32_t tag;      // NCCL isend tag, echoed so the receiver can verify
+                    // seq-ordinal matching against the posted irecv tag
+                    // (a silent cross-match becomes a loud EPROTO)
 };
-static_assert(sizeof(ChunkHdr) == 16, "wire layout");
+static_assert(sizeof(ChunkHdr) == 20, "wire layout");
 
 // NCCL handle (<= NCCL_NET_HANDLE_MAXSIZE = 128 bytes).  `stage` stashes the
 // connector-side in-progress state across nonblocking connect() retries
@@ -81,7 +84,8 @@ struct ListenHandle {
   uint16_t family;  // AF_INET / AF_INET6
   uint16_t port;    // network byte order
   uint8_t addr[16];
-  uint64_t stage;
+  uint64_t stage;        // ConnectTask* — validated against a live-task
+  uint64_t stage_token;  // registry + this conn_id before any dereference
 };
 static_assert(sizeof(ListenHandle) <= NCCL_NET_HANDLE_MAXSIZE, "handle size");
 
@@ -133,6 +137,8 @@ struct SendRequest {
   std::atomic<uint32_t> avail{0};   // staged watermark; == total for host src
   std::atomic<uint32_t> sent{0};    // bytes fully handed to the kernel
   std::atomic<bool> hdr_sent{false};
+  int tag = 0;  // NCCL tag, echoed in every ChunkHdr (written before the
+                // state_seq publish; read only after a confirmed claim)
   StageAlloc* stage = nullptr;  // non-null for NCCL_PTR_CUDA sends
   SendComm* comm = nullptr;
   uint32_t span_slot = UINT32_MAX;
@@ -148,6 +154,7 @@ struct RecvRequest {
   std::atomic<uint64_t> state_seq{pack_ss(UINT32_MAX, REQ_FREE)};
   char* dst = nullptr;     // user destination (host) — staging writes here
   uint32_t capacity = 0;   // posted buffer size (recv may be smaller)
+  int tag = 0;             // posted NCCL tag; checked against ChunkHdr.tag
   std::atomic<int64_t> total{-1};      // from first chunk header
   std::atomic<uint32_t> received{0};   // socket bytes landed
   std::atomic<bool> gpu_done{false};   // H2D staging drained (CUDA dst)
@@ -291,6 +298,7 @@ struct RecvComm {
 struct ConnectTask {  // connector side, lives in ListenHandle::stage
   int dev = 0;
   uint64_t conn_id = 0;
+  uint64_t last_touch_ns = 0;  // abandoned-task reaping (connect registry)
   sockaddr_storage peer{};
   socklen_t peer_len = 0;
   struct Pending {

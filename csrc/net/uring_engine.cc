@@ -230,9 +230,9 @@ class UringIoThread : public IIoThread {
         int i = s->ur.nchunks++;
         s->ur.reqs[i] = r;
         s->ur.hdrs[i] = {ss_seq(r->state_seq.load(std::memory_order_relaxed)),
-                         off, len, r->total};
+                         off, len, r->total, r->tag};
         s->ur.payloads[i] = r->src + off;
-        s->ur.batch_bytes += 16 + len;
+        s->ur.batch_bytes += (uint32_t)sizeof(ChunkHdr) + len;
       }
       if (s->ur.nchunks == 0) return;  // idle (snd_idle set) until kicked
     s->snd_idle.store(false, std::memory_order_relaxed);
@@ -242,16 +242,17 @@ class UringIoThread : public IIoThread {
     int iovn = 0;
     uint32_t skip = s->ur.done;
     for (int i = 0; i < s->ur.nchunks; i++) {
-      uint32_t span = 16 + s->ur.hdrs[i].len;
+      uint32_t span = (uint32_t)sizeof(ChunkHdr) + s->ur.hdrs[i].len;
       if (skip >= span) {
         skip -= span;
         continue;
       }
       uint32_t o = skip;
       skip = 0;
-      if (o < 16)
-        s->ur.iov[iovn++] = {(char*)&s->ur.hdrs[i] + o, 16 - o};
-      uint32_t pay_off = o > 16 ? o - 16 : 0;
+      if (o < sizeof(ChunkHdr))
+        s->ur.iov[iovn++] = {(char*)&s->ur.hdrs[i] + o, sizeof(ChunkHdr) - o};
+      uint32_t pay_off =
+          o > sizeof(ChunkHdr) ? o - (uint32_t)sizeof(ChunkHdr) : 0;
       if (s->ur.hdrs[i].len > pay_off)
         s->ur.iov[iovn++] = {(void*)(s->ur.payloads[i] + pay_off),
                              s->ur.hdrs[i].len - pay_off};
@@ -307,16 +308,17 @@ class UringIoThread : public IIoThread {
     int iovn = 0;
     uint32_t skip = s->ur.done;
     for (int i = 0; i < s->ur.nchunks; i++) {
-      uint32_t span = 16 + s->ur.hdrs[i].len;
+      uint32_t span = (uint32_t)sizeof(ChunkHdr) + s->ur.hdrs[i].len;
       if (skip >= span) {
         skip -= span;
         continue;
       }
       uint32_t o = skip;
       skip = 0;
-      if (o < 16)
-        s->ur.iov[iovn++] = {(char*)&s->ur.hdrs[i] + o, 16 - o};
-      uint32_t pay_off = o > 16 ? o - 16 : 0;
+      if (o < sizeof(ChunkHdr))
+        s->ur.iov[iovn++] = {(char*)&s->ur.hdrs[i] + o, sizeof(ChunkHdr) - o};
+      uint32_t pay_off =
+          o > sizeof(ChunkHdr) ? o - (uint32_t)sizeof(ChunkHdr) : 0;
       if (s->ur.hdrs[i].len > pay_off)
         s->ur.iov[iovn++] = {(void*)(s->ur.payloads[i] + pay_off),
                              s->ur.hdrs[i].len - pay_off};

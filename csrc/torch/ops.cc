@@ -39,11 +39,17 @@ hipStream_t current_stream() {
   return (hipStream_t)c10::cuda::getCurrentCUDAStream().stream();
 }
 
-// Persistent per-process scratch for descriptor tables.
+// Per-launch descriptor-table buffers, in a ring: a single shared pinned
+// buffer could be overwritten by the CPU while a previous launch's
+// hipMemcpyAsync of it was still queued (the GPU runs behind the CPU) —
+// torn descriptor tables, silent corruption.  Each slot carries an event
+// recorded after its launch; a slot is reused only after that event lands.
 struct Scratch {
   void* dev = nullptr;
   void* host = nullptr;
   size_t size = 0;
+  hipEvent_t ev = nullptr;
+  bool busy = false;
   void ensure(size_t need) {
     if (need <= size) return;
     size_t sz = std::max(need, (size_t)64 * 1024);
@@ -56,8 +62,25 @@ struct Scratch {
   }
 };
 Scratch& scratch() {
-  static Scratch s;
+  static Scratch ring[4];
+  static int next = 0;
+  Scratch& s = ring[next];
+  next = (next + 1) % 4;
+  if (s.busy) {
+    TORCH_CHECK(hipEventSynchronize(s.ev) == hipSuccess, "scratch event");
+    s.busy = false;
+  }
   return s;
+}
+
+// Record the in-flight marker after the launch that consumed `s`.
+void scratch_commit(Scratch& s, hipStream_t stream) {
+  if (!s.ev)
+    TORCH_CHECK(
+        hipEventCreateWithFlags(&s.ev, hipEventDisableTiming) == hipSuccess,
+        "scratch event create");
+  TORCH_CHECK(hipEventRecord(s.ev, stream) == hipSuccess, "scratch record");
+  s.busy = true;
 }
 
 void copy_bytes(torch::Tensor dst, torch::Tensor src) {
@@ -80,6 +103,8 @@ void build_descs(torch::Tensor& flat, std::vector<torch::Tensor>& tensors,
                 "bucket tensors must be contiguous CUDA tensors");
     TORCH_CHECK(t.scalar_type() == flat.scalar_type(), "dtype mismatch");
     int64_t nb = t.nbytes();
+    TORCH_CHECK(nb <= (int64_t)UINT32_MAX,
+                "tensor too large for 32-bit descriptor (", nb, " bytes)");
     TORCH_CHECK(off + nb <= (int64_t)flat.nbytes(), "flat too small");
     if (pack)
       descs.push_back({t.data_ptr(), base + off, (uint32_t)nb});
@@ -99,6 +124,7 @@ void run_multi(std::vector<PackDesc>& descs) {
                              s.host, current_stream());
     TORCH_CHECK(need == 0, "multi_copy_launch failed (", need, " bytes)");
   }
+  scratch_commit(s, current_stream());
 }
 
 // Pack `tensors` back-to-back into `flat` (one fused kernel).
@@ -138,6 +164,8 @@ void fused_sgd(std::vector<torch::Tensor> params,
                 "contiguous tensors required");
     TORCH_CHECK(p.numel() == g.numel() && p.numel() == m.numel(),
                 "numel mismatch");
+    TORCH_CHECK(p.numel() <= (int64_t)UINT32_MAX,
+                "tensor too large for 32-bit descriptor");
     descs.push_back({p.data_ptr<float>(), g.data_ptr<float>(),
                      m.data_ptr<float>(), (uint32_t)p.numel()});
   }
@@ -154,6 +182,7 @@ void fused_sgd(std::vector<torch::Tensor> params,
                             current_stream());
     TORCH_CHECK(need == 0, "multi_sgd_launch failed");
   }
+  scratch_commit(s, current_stream());
 }
 
 }  // namespace

@@ -230,7 +230,7 @@ def _malformed_frames(env, q):
     port = socket.ntohs(port_be)
     # fake a single-stream sender: valid hello, then a malformed chunk
     conn_id = 0xDEADBEEFCAFEF00D
-    hello = struct.pack("<IIQHHI", magic, 1, conn_id, 0, 1, 0)
+    hello = struct.pack("<IIQHHI", magic, 2, conn_id, 0, 1, 0)
     g = socket.create_connection(("127.0.0.1", port), timeout=5)
     g.sendall(hello)
     rcomm = None
@@ -243,7 +243,7 @@ def _malformed_frames(env, q):
     req = p.irecv(rcomm, buf, 4096, mh)
     assert req is not None
     # header with len > total — must surface an error, not hang
-    g.sendall(struct.pack("<IIII", 0, 0, 4096, 16))
+    g.sendall(struct.pack("<IIIIi", 0, 0, 4096, 16, 0))
     t0 = time.monotonic()
     while time.monotonic() - t0 < 30:
         try:
@@ -669,3 +669,84 @@ def test_ipv6_family():
            "BNET_NSTREAMS": "3", "BNET_MIN_CHUNKSIZE": "8192"}
     res = _run_sub(_ipv6_roundtrip, env)
     assert res in ("ok", "skip")
+
+
+def _tag_roundtrip(env, q):
+    for k, v in env.items():
+        os.environ[k] = v
+    import ctypes as C
+    import time
+
+    from baguanet.plugin import Plugin
+
+    p = Plugin()
+    handle, lcomm = p.listen(0)
+    scomm = rcomm = None
+    t0 = time.monotonic()
+    while scomm is None or rcomm is None:
+        assert time.monotonic() - t0 < 30
+        if scomm is None:
+            scomm = p.connect(0, handle)
+        if rcomm is None:
+            rcomm = p.accept(lcomm)
+    mh = p.reg_mr(scomm, None, 0)
+    # matching tags: completes normally
+    buf = C.create_string_buffer(b"tagged-payload-ok", 17)
+    rbuf = C.create_string_buffer(17)
+    rreq = p.irecv(rcomm, rbuf, 17, mh, tag=42)
+    sreq = p.isend(scomm, buf, 17, mh, tag=42)
+    assert p.wait(sreq, 30) == 17 and p.wait(rreq, 30) == 17
+    assert rbuf.raw[:17] == b"tagged-payload-ok"
+    # mismatched tags: the echoed tag turns a silent cross-match into a
+    # loud EPROTO on the recv comm
+    rreq = p.irecv(rcomm, rbuf, 17, mh, tag=9)
+    sreq = p.isend(scomm, buf, 17, mh, tag=7)
+    t0 = time.monotonic()
+    while time.monotonic() - t0 < 30:
+        try:
+            done, _ = p.test(rreq)
+        except RuntimeError:
+            q.put("errored")
+            return
+        if done:
+            q.put("unexpected-done")
+            return
+        time.sleep(0.01)
+    q.put("timeout")
+
+
+def test_tag_echo_mismatch_is_loud():
+    """isend tags are echoed on the wire; a cross-match errors the comm."""
+    assert _run_sub(_tag_roundtrip, {"NCCL_SOCKET_IFNAME": "lo"}) == "errored"
+
+
+def _touch_plugin(env, q):
+    for k, v in env.items():
+        os.environ[k] = v
+    from baguanet.plugin import Plugin
+
+    Plugin()  # init reads config; atexit dumps telemetry
+    q.put("ok")
+
+
+def test_rank_templated_telemetry_files(tmp_path):
+    """N-rank jobs must leave N distinguishable metric files:
+    RANK is appended as .r<rank> (or substituted for %r), and counters
+    carry a rank label."""
+    base = str(tmp_path / "m.prom")
+    assert _run_sub(_touch_plugin, {
+        "NCCL_SOCKET_IFNAME": "lo",
+        "RANK": "3",
+        "BNET_METRICS_FILE": base,
+    }) == "ok"
+    f = tmp_path / "m.prom.r3"
+    assert f.exists(), "expected rank-suffixed metrics file"
+    assert 'bnet_isend_total{rank="3"}' in f.read_text()
+
+    templ = str(tmp_path / "rank-%r.prom")
+    assert _run_sub(_touch_plugin, {
+        "NCCL_SOCKET_IFNAME": "lo",
+        "BNET_RANK": "5",
+        "BNET_METRICS_FILE": templ,
+    }) == "ok"
+    assert (tmp_path / "rank-5.prom").exists()
