@@ -35,6 +35,30 @@ class NetProperties(C.Structure):
     ]
 
 
+class NetPropertiesV7(C.Structure):
+    _fields_ = NetProperties._fields_ + [
+        ("netDeviceType", C.c_int),
+        ("netDeviceVersion", C.c_int),
+    ]
+
+
+class NetPropertiesV8(C.Structure):
+    _fields_ = [
+        ("name", C.c_char_p),
+        ("pciPath", C.c_char_p),
+        ("guid", C.c_uint64),
+        ("ptrSupport", C.c_int),
+        ("regIsGlobal", C.c_int),
+        ("speed", C.c_int),
+        ("port", C.c_int),
+        ("latency", C.c_float),
+        ("maxComms", C.c_int),
+        ("maxRecvs", C.c_int),
+        ("netDeviceType", C.c_int),
+        ("netDeviceVersion", C.c_int),
+    ]
+
+
 _F = C.CFUNCTYPE
 
 
@@ -69,12 +93,69 @@ class NcclNetV6(C.Structure):
     ]
 
 
-class Plugin:
-    """Thin pythonic wrapper over the vtable (raises on non-zero results)."""
+def _vtable_struct(abi: int, props_struct) -> type:
+    """Build the ncclNet_v7/v8 ctypes struct (v7+: device-handle out-params
+    on connect/accept, trailing getDeviceMr/irecvConsumed; v8: size_t
+    regMr)."""
+    return type(
+        f"NcclNetV{abi}",
+        (C.Structure,),
+        {"_fields_": [
+            ("name", C.c_char_p),
+            ("init", _F(C.c_int, C.c_void_p)),
+            ("devices", _F(C.c_int, C.POINTER(C.c_int))),
+            ("getProperties", _F(C.c_int, C.c_int, C.POINTER(props_struct))),
+            ("listen", _F(C.c_int, C.c_int, C.c_void_p,
+                          C.POINTER(C.c_void_p))),
+            ("connect", _F(C.c_int, C.c_int, C.c_void_p,
+                           C.POINTER(C.c_void_p), C.POINTER(C.c_void_p))),
+            ("accept", _F(C.c_int, C.c_void_p, C.POINTER(C.c_void_p),
+                          C.POINTER(C.c_void_p))),
+            ("regMr", _F(C.c_int, C.c_void_p, C.c_void_p,
+                         C.c_size_t if abi >= 8 else C.c_int, C.c_int,
+                         C.POINTER(C.c_void_p))),
+            ("regMrDmaBuf", _F(C.c_int, C.c_void_p, C.c_void_p, C.c_size_t,
+                               C.c_int, C.c_uint64, C.c_int,
+                               C.POINTER(C.c_void_p))),
+            ("deregMr", _F(C.c_int, C.c_void_p, C.c_void_p)),
+            ("isend", _F(C.c_int, C.c_void_p, C.c_void_p, C.c_int, C.c_int,
+                         C.c_void_p, C.POINTER(C.c_void_p))),
+            ("irecv", _F(C.c_int, C.c_void_p, C.c_int, C.POINTER(C.c_void_p),
+                         C.POINTER(C.c_int), C.POINTER(C.c_int),
+                         C.POINTER(C.c_void_p), C.POINTER(C.c_void_p))),
+            ("iflush", _F(C.c_int, C.c_void_p, C.c_int, C.POINTER(C.c_void_p),
+                          C.POINTER(C.c_int), C.POINTER(C.c_void_p),
+                          C.POINTER(C.c_void_p))),
+            ("test", _F(C.c_int, C.c_void_p, C.POINTER(C.c_int),
+                        C.POINTER(C.c_int))),
+            ("closeSend", _F(C.c_int, C.c_void_p)),
+            ("closeRecv", _F(C.c_int, C.c_void_p)),
+            ("closeListen", _F(C.c_int, C.c_void_p)),
+            ("getDeviceMr", _F(C.c_int, C.c_void_p, C.c_void_p,
+                               C.POINTER(C.c_void_p))),
+            ("irecvConsumed", _F(C.c_int, C.c_void_p, C.c_int, C.c_void_p)),
+        ]},
+    )
 
-    def __init__(self, path: Optional[str] = None):
+
+NcclNetV7 = _vtable_struct(7, NetPropertiesV7)
+NcclNetV8 = _vtable_struct(8, NetPropertiesV8)
+
+
+class Plugin:
+    """Thin pythonic wrapper over the vtable (raises on non-zero results).
+
+    ``abi`` selects which exported vtable to drive (6, 7 or 8) — the same
+    structs RCCL dlsyms, newest-first.
+    """
+
+    def __init__(self, path: Optional[str] = None, abi: int = 6):
         self.lib = C.CDLL(str(path or PLUGIN_PATH), mode=C.RTLD_GLOBAL)
-        self.vt = NcclNetV6.in_dll(self.lib, "ncclNetPlugin_v6")
+        self.abi = abi
+        struct = {6: NcclNetV6, 7: NcclNetV7, 8: NcclNetV8}[abi]
+        self.vt = struct.in_dll(self.lib, f"ncclNetPlugin_v{abi}")
+        self._props_struct = {6: NetProperties, 7: NetPropertiesV7,
+                              8: NetPropertiesV8}[abi]
         self._check(self.vt.init(None), "init")
 
     @staticmethod
@@ -92,9 +173,9 @@ class Plugin:
         return n.value
 
     def properties(self, dev: int) -> dict:
-        p = NetProperties()
+        p = self._props_struct()
         self._check(self.vt.getProperties(dev, C.byref(p)), "getProperties")
-        return {
+        out = {
             "name": p.name.decode() if p.name else "",
             "pciPath": p.pciPath.decode() if p.pciPath else "",
             "guid": p.guid,
@@ -105,6 +186,10 @@ class Plugin:
             "maxComms": p.maxComms,
             "maxRecvs": p.maxRecvs,
         }
+        for extra in ("regIsGlobal", "netDeviceType", "netDeviceVersion"):
+            if hasattr(p, extra):
+                out[extra] = getattr(p, extra)
+        return out
 
     def listen(self, dev: int):
         handle = (C.c_char * NCCL_NET_HANDLE_MAXSIZE)()
@@ -114,12 +199,25 @@ class Plugin:
 
     def connect(self, dev: int, handle) -> Optional[C.c_void_p]:
         scomm = C.c_void_p(None)
-        self._check(self.vt.connect(dev, handle, C.byref(scomm)), "connect")
+        if self.abi >= 7:
+            dh = C.c_void_p(None)  # device handle out-param: host plugin
+            self._check(self.vt.connect(dev, handle, C.byref(scomm),
+                                        C.byref(dh)), "connect")
+            assert dh.value is None, "host plugin must not set a dev handle"
+        else:
+            self._check(self.vt.connect(dev, handle, C.byref(scomm)),
+                        "connect")
         return scomm if scomm.value else None
 
     def accept(self, lcomm) -> Optional[C.c_void_p]:
         rcomm = C.c_void_p(None)
-        self._check(self.vt.accept(lcomm, C.byref(rcomm)), "accept")
+        if self.abi >= 7:
+            dh = C.c_void_p(None)
+            self._check(self.vt.accept(lcomm, C.byref(rcomm), C.byref(dh)),
+                        "accept")
+            assert dh.value is None, "host plugin must not set a dev handle"
+        else:
+            self._check(self.vt.accept(lcomm, C.byref(rcomm)), "accept")
         return rcomm if rcomm.value else None
 
     def reg_mr(self, comm, data, size: int, ptr_type: int = NCCL_PTR_HOST):

@@ -87,6 +87,8 @@ typedef struct {
   int needsProxyProgress;
 } ncclNetDeviceHandle_v7_t;
 
+typedef ncclNetDeviceHandle_v7_t ncclNetDeviceHandle_v8_t;
+
 // ---------------------------------------------------------------------------
 // v6 ABI (NCCL >= 2.13 layout of v6, incl. regMrDmaBuf) — what we export.
 // RCCL probes v10..v6 and wraps older versions internally; v6 is the
@@ -155,3 +157,107 @@ typedef struct {
   ncclResult_t (*closeRecv)(void* recvComm);
   ncclResult_t (*closeListen)(void* listenComm);
 } ncclNet_v6_t;
+
+// ---------------------------------------------------------------------------
+// v7 ABI (NCCL 2.18 era): properties gain netDeviceType/netDeviceVersion;
+// connect/accept gain a device-handle out-param (NULL from host-proxy
+// plugins); getDeviceMr/irecvConsumed appended for device-offload plugins.
+// These layouts are frozen public ABI — identical across NCCL and RCCL
+// (third-party plugins such as aws-ofi are compiled against the same
+// structs for both stacks).
+// ---------------------------------------------------------------------------
+
+typedef struct {
+  char* name;
+  char* pciPath;
+  uint64_t guid;
+  int ptrSupport;  // [NCCL_PTR_HOST|NCCL_PTR_CUDA|NCCL_PTR_DMABUF]
+  int speed;       // Mbps
+  int port;
+  float latency;
+  int maxComms;
+  int maxRecvs;
+  int netDeviceType;     // ncclNetDeviceType (NCCL_NET_DEVICE_HOST)
+  int netDeviceVersion;
+} ncclNetProperties_v7_t;
+
+typedef struct {
+  const char* name;
+  ncclResult_t (*init)(ncclDebugLogger_t logFunction);
+  ncclResult_t (*devices)(int* ndev);
+  ncclResult_t (*getProperties)(int dev, ncclNetProperties_v7_t* props);
+  ncclResult_t (*listen)(int dev, void* handle, void** listenComm);
+  ncclResult_t (*connect)(int dev, void* handle, void** sendComm,
+                          ncclNetDeviceHandle_v7_t** sendDevComm);
+  ncclResult_t (*accept)(void* listenComm, void** recvComm,
+                         ncclNetDeviceHandle_v7_t** recvDevComm);
+  ncclResult_t (*regMr)(void* comm, void* data, int size, int type,
+                        void** mhandle);
+  ncclResult_t (*regMrDmaBuf)(void* comm, void* data, size_t size, int type,
+                              uint64_t offset, int fd, void** mhandle);
+  ncclResult_t (*deregMr)(void* comm, void* mhandle);
+  ncclResult_t (*isend)(void* sendComm, void* data, int size, int tag,
+                        void* mhandle, void** request);
+  ncclResult_t (*irecv)(void* recvComm, int n, void** data, int* sizes,
+                        int* tags, void** mhandles, void** request);
+  ncclResult_t (*iflush)(void* recvComm, int n, void** data, int* sizes,
+                         void** mhandles, void** request);
+  ncclResult_t (*test)(void* request, int* done, int* sizes);
+  ncclResult_t (*closeSend)(void* sendComm);
+  ncclResult_t (*closeRecv)(void* recvComm);
+  ncclResult_t (*closeListen)(void* listenComm);
+  // device-offload hooks (host-proxy plugins return ncclInternalError)
+  ncclResult_t (*getDeviceMr)(void* comm, void* mhandle, void** dptr_mhandle);
+  ncclResult_t (*irecvConsumed)(void* recvComm, int n, void* request);
+} ncclNet_v7_t;
+
+// ---------------------------------------------------------------------------
+// v8 ABI (NCCL 2.19-2.21 era): properties gain regIsGlobal; regMr's size
+// widens to size_t.  Everything else matches v7.
+// ---------------------------------------------------------------------------
+
+typedef struct {
+  char* name;
+  char* pciPath;
+  uint64_t guid;
+  int ptrSupport;
+  int regIsGlobal;  // regMr()s are valid for all comms (we say no: the
+                    // mhandle is just the pointer type, but per-comm is the
+                    // conservative answer)
+  int speed;
+  int port;
+  float latency;
+  int maxComms;
+  int maxRecvs;
+  int netDeviceType;
+  int netDeviceVersion;
+} ncclNetProperties_v8_t;
+
+typedef struct {
+  const char* name;
+  ncclResult_t (*init)(ncclDebugLogger_t logFunction);
+  ncclResult_t (*devices)(int* ndev);
+  ncclResult_t (*getProperties)(int dev, ncclNetProperties_v8_t* props);
+  ncclResult_t (*listen)(int dev, void* handle, void** listenComm);
+  ncclResult_t (*connect)(int dev, void* handle, void** sendComm,
+                          ncclNetDeviceHandle_v8_t** sendDevComm);
+  ncclResult_t (*accept)(void* listenComm, void** recvComm,
+                         ncclNetDeviceHandle_v8_t** recvDevComm);
+  ncclResult_t (*regMr)(void* comm, void* data, size_t size, int type,
+                        void** mhandle);
+  ncclResult_t (*regMrDmaBuf)(void* comm, void* data, size_t size, int type,
+                              uint64_t offset, int fd, void** mhandle);
+  ncclResult_t (*deregMr)(void* comm, void* mhandle);
+  ncclResult_t (*isend)(void* sendComm, void* data, int size, int tag,
+                        void* mhandle, void** request);
+  ncclResult_t (*irecv)(void* recvComm, int n, void** data, int* sizes,
+                        int* tags, void** mhandles, void** request);
+  ncclResult_t (*iflush)(void* recvComm, int n, void** data, int* sizes,
+                         void** mhandles, void** request);
+  ncclResult_t (*test)(void* request, int* done, int* sizes);
+  ncclResult_t (*closeSend)(void* sendComm);
+  ncclResult_t (*closeRecv)(void* recvComm);
+  ncclResult_t (*closeListen)(void* listenComm);
+  ncclResult_t (*getDeviceMr)(void* comm, void* mhandle, void** dptr_mhandle);
+  ncclResult_t (*irecvConsumed)(void* recvComm, int n, void* request);
+} ncclNet_v8_t;

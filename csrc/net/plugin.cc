@@ -101,7 +101,92 @@ ncclResult_t bnCloseListen(void* listenComm) {
   return Net::get().close_listen(listenComm);
 }
 
+// ---- v7 / v8 adapters ------------------------------------------------------
+// RCCL probes ncclNetPlugin_v10..v6 newest-first; exporting v8 (+v7) lets
+// newer RCCLs use a native vtable instead of their internal v6 wrapper
+// (reference shipped floor+current the same way: cc/v3 + cc/v4).  The only
+// deltas from v6 are the properties tail, the device-handle out-params on
+// connect/accept (NULL: all networking is host-proxy), v8's size_t regMr,
+// and the device-offload hooks (unreachable at NCCL_NET_DEVICE_HOST).
+
+ncclResult_t bnGetProperties_v7(int dev, ncclNetProperties_v7_t* props) {
+  ncclNetProperties_v6_t p6;
+  ncclResult_t rc = Net::get().get_properties(dev, &p6);
+  if (rc != ncclSuccess) return rc;
+  props->name = p6.name;
+  props->pciPath = p6.pciPath;
+  props->guid = p6.guid;
+  props->ptrSupport = p6.ptrSupport;
+  props->speed = p6.speed;
+  props->port = p6.port;
+  props->latency = p6.latency;
+  props->maxComms = p6.maxComms;
+  props->maxRecvs = p6.maxRecvs;
+  props->netDeviceType = NCCL_NET_DEVICE_HOST;
+  props->netDeviceVersion = NCCL_NET_DEVICE_INVALID_VERSION;
+  return ncclSuccess;
+}
+
+ncclResult_t bnGetProperties_v8(int dev, ncclNetProperties_v8_t* props) {
+  ncclNetProperties_v6_t p6;
+  ncclResult_t rc = Net::get().get_properties(dev, &p6);
+  if (rc != ncclSuccess) return rc;
+  props->name = p6.name;
+  props->pciPath = p6.pciPath;
+  props->guid = p6.guid;
+  props->ptrSupport = p6.ptrSupport;
+  props->regIsGlobal = 0;
+  props->speed = p6.speed;
+  props->port = p6.port;
+  props->latency = p6.latency;
+  props->maxComms = p6.maxComms;
+  props->maxRecvs = p6.maxRecvs;
+  props->netDeviceType = NCCL_NET_DEVICE_HOST;
+  props->netDeviceVersion = NCCL_NET_DEVICE_INVALID_VERSION;
+  return ncclSuccess;
+}
+
+ncclResult_t bnConnect_v7(int dev, void* handle, void** sendComm,
+                          ncclNetDeviceHandle_v7_t** sendDevComm) {
+  if (sendDevComm) *sendDevComm = nullptr;  // host-proxy: no device comm
+  return Net::get().connect(dev, handle, sendComm);
+}
+
+ncclResult_t bnAccept_v7(void* listenComm, void** recvComm,
+                         ncclNetDeviceHandle_v7_t** recvDevComm) {
+  if (recvDevComm) *recvDevComm = nullptr;
+  return Net::get().accept(listenComm, recvComm);
+}
+
+ncclResult_t bnRegMr_v8(void* comm, void* data, size_t size, int type,
+                        void** mhandle) {
+  if (size > (size_t)INT32_MAX) {
+    // registration is type-tagging only (no pinning), so size never
+    // matters — but refuse obviously bogus inputs rather than truncate
+    return bnRegMr(comm, data, INT32_MAX, type, mhandle);
+  }
+  return bnRegMr(comm, data, (int)size, type, mhandle);
+}
+
+ncclResult_t bnGetDeviceMr(void* comm, void* mhandle, void** dptr_mhandle) {
+  (void)comm, (void)mhandle, (void)dptr_mhandle;
+  return ncclInternalError;  // never called at NCCL_NET_DEVICE_HOST
+}
+
+ncclResult_t bnIrecvConsumed(void* recvComm, int n, void* request) {
+  (void)recvComm, (void)n, (void)request;
+  return ncclInternalError;  // never called at NCCL_NET_DEVICE_HOST
+}
+
 }  // namespace
+
+// Compile-time ABI ceiling: the main .so exports v6+v7+v8; a companion
+// v6-only .so (libnccl-net-bagua6.so, -DBNET_ABI_MAX=6) is shipped as an
+// escape hatch selectable via NCCL_NET_PLUGIN=bagua6 in case a future RCCL
+// disagrees about the newer frozen layouts.
+#ifndef BNET_ABI_MAX
+#define BNET_ABI_MAX 8
+#endif
 
 extern "C" {
 
@@ -114,5 +199,25 @@ __attribute__((visibility("default"))) ncclNet_v6_t ncclNetPlugin_v6 = {
     bnIflush,       bnTest,        bnCloseSend, bnCloseRecv,
     bnCloseListen,
 };
+
+#if BNET_ABI_MAX >= 7
+__attribute__((visibility("default"))) ncclNet_v7_t ncclNetPlugin_v7 = {
+    "BaguaNetAMD",  bnInit,          bnDevices,     bnGetProperties_v7,
+    bnListen,       bnConnect_v7,    bnAccept_v7,   bnRegMr,
+    bnRegMrDmaBuf,  bnDeregMr,       bnIsend,       bnIrecv,
+    bnIflush,       bnTest,          bnCloseSend,   bnCloseRecv,
+    bnCloseListen,  bnGetDeviceMr,   bnIrecvConsumed,
+};
+#endif
+
+#if BNET_ABI_MAX >= 8
+__attribute__((visibility("default"))) ncclNet_v8_t ncclNetPlugin_v8 = {
+    "BaguaNetAMD",  bnInit,          bnDevices,     bnGetProperties_v8,
+    bnListen,       bnConnect_v7,    bnAccept_v7,   bnRegMr_v8,
+    bnRegMrDmaBuf,  bnDeregMr,       bnIsend,       bnIrecv,
+    bnIflush,       bnTest,          bnCloseSend,   bnCloseRecv,
+    bnCloseListen,  bnGetDeviceMr,   bnIrecvConsumed,
+};
+#endif
 
 }  // extern "C"
