@@ -18,6 +18,8 @@ from __future__ import annotations
 import argparse
 import json
 import os
+import socket
+import subprocess
 import sys
 import time
 
@@ -27,6 +29,83 @@ sys.path.insert(0, REPO)
 # Reference numbers: VGG16 synthetic on 4x8xV100 100GbE,
 # 126.5 img/sec/GPU with bagua-net (4046.6 total at 32 GPUs).
 BASELINE_PER_GPU = 126.5
+
+
+def _free_port() -> int:
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def _clean_env() -> dict:
+    """Fresh env for a sub-torchrun: drop the parent's elastic/NCCL state."""
+    drop_exact = {"RANK", "LOCAL_RANK", "WORLD_SIZE", "LOCAL_WORLD_SIZE",
+                  "GROUP_RANK", "GROUP_WORLD_SIZE", "ROLE_RANK",
+                  "ROLE_WORLD_SIZE", "ROLE_NAME", "MASTER_ADDR",
+                  "MASTER_PORT", "OMP_NUM_THREADS"}
+    drop_prefix = ("NCCL_", "RCCL_", "TORCHELASTIC", "PET_", "BNET_")
+    return {k: v for k, v in os.environ.items()
+            if k not in drop_exact and not k.startswith(drop_prefix)}
+
+
+def run_busbw_ab(world: int, iters: int, use_cuda: bool,
+                 max_bytes: int = 128 * 1024 * 1024) -> dict:
+    """The BASELINE headline A/B: all_reduce_perf busbw 8B-128M through the
+    plugin (--force-net) vs stock RCCL TCP (--no-plugin --force-net) — the
+    exact +50%-over-stock comparison the reference published (reference
+    README.md:27-50).  Runs two sub-torchruns at the same world size;
+    returns {sizes, plugin, stock, ratio} busbw arrays (GB/s)."""
+    out = {"ab_status": "ok"}
+    variants = {"plugin": ["--force-net"],
+                "stock": ["--no-plugin", "--force-net"]}
+    tables = {}
+    for name, flags in variants.items():
+        res_file = os.path.join(REPO, f".ab_{name}_{os.getpid()}.json")
+        cmd = [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", f"--nproc-per-node={world}",
+            "--master-addr", "127.0.0.1",
+            "--master-port", str(_free_port()),
+            os.path.join(REPO, "benchmarks", "allreduce_perf.py"),
+            "--max-bytes", str(max_bytes),
+            "--iters", str(iters), "--warmup", "3",
+            "--out", res_file, *flags,
+        ]
+        env = _clean_env()  # allreduce_perf auto-selects gloo without CUDA
+        try:
+            p = subprocess.run(cmd, env=env, capture_output=True, text=True,
+                               timeout=600)
+            if p.returncode != 0:
+                out["ab_status"] = (
+                    f"{name} failed rc={p.returncode}: "
+                    + p.stderr.strip().splitlines()[-1][:200]
+                    if p.stderr.strip() else f"{name} failed"
+                )
+                return out
+            with open(res_file) as f:
+                tables[name] = json.load(f)["results"]
+        except (subprocess.TimeoutExpired, OSError, KeyError,
+                json.JSONDecodeError, IndexError) as e:
+            out["ab_status"] = f"{name} error: {type(e).__name__} {e}"[:200]
+            return out
+        finally:
+            try:
+                os.unlink(res_file)
+            except OSError:
+                pass
+    sizes = [r["bytes"] for r in tables["plugin"]]
+    plugin = [r["busbw_GBps"] for r in tables["plugin"]]
+    stock = [r["busbw_GBps"] for r in tables["stock"]]
+    out["sizes"] = sizes
+    out["busbw_plugin_GBps"] = plugin
+    out["busbw_stock_GBps"] = stock[:len(plugin)]
+    out["ratio"] = [
+        round(p / s, 3) if s else None
+        for p, s in zip(plugin, out["busbw_stock_GBps"])
+    ]
+    return out
 
 
 def main():
@@ -44,6 +123,12 @@ def main():
                     default=True,
                     help="NHWC memory format (MIOpen igemm path; +9%% on "
                          "VGG16 fp32, same numerics)")
+    ap.add_argument("--ab", choices=["auto", "always", "never"],
+                    default="auto",
+                    help="plugin-vs-stock all_reduce_perf busbw companion "
+                         "sweep after the timed run (auto: when world>1)")
+    ap.add_argument("--ab-iters", type=int, default=10)
+    ap.add_argument("--ab-max-bytes", type=int, default=128 * 1024 * 1024)
     args = ap.parse_args()
 
     # plugin env must be set before the first collective; preload the .so
@@ -135,6 +220,24 @@ def main():
     ms_per_step = elapsed / args.steps * 1000.0
     total_img_per_sec = args.batch * n_gpus * args.steps / elapsed
 
+    if world > 1:
+        dist.destroy_process_group()
+    if rank != 0:
+        return
+
+    # Plugin-engaged companion (at N>1 the bench must report a
+    # plugin-engaged number): after the DDP ranks wind down, rank 0
+    # sub-launches the busbw A/B at the same world size.  The timed metric
+    # above is untouched — this runs outside the timed region.
+    companion = None
+    want_ab = args.ab == "always" or (args.ab == "auto" and world > 1
+                                      and use_cuda)
+    if want_ab:
+        if world > 1:
+            time.sleep(5.0)  # let sibling ranks exit and release their GPUs
+        companion = run_busbw_ab(max(world, 1), args.ab_iters, use_cuda,
+                                 args.ab_max_bytes)
+
     if rank == 0:
         out = {
             "metric": "VGG16 synthetic training img/sec (total)"
@@ -163,10 +266,11 @@ def main():
                 "(4x8xV100 100GbE, BASELINE.md) scaled to n_gpus",
             },
         }
+        if companion is not None:
+            # the BASELINE headline's other half: all_reduce_perf busbw
+            # 8B-128M, plugin vs stock TCP, at this GPU count
+            out["companion_busbw_ab"] = companion
         print(json.dumps(out))
-
-    if world > 1:
-        dist.destroy_process_group()
 
 
 if __name__ == "__main__":

@@ -56,6 +56,10 @@ def main():
     elif args.force_net:
         os.environ["NCCL_P2P_DISABLE"] = "1"
         os.environ["NCCL_SHM_DISABLE"] = "1"
+        # pin the internal TCP socket transport: "stock RCCL TCP" is the
+        # comparison the reference published (README.md:48-50), and on an
+        # RDMA-equipped box RCCL would otherwise pick IB verbs
+        os.environ.setdefault("NCCL_NET", "Socket")
 
     import torch
     import torch.distributed as dist

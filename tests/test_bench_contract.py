@@ -73,3 +73,26 @@ def test_allreduce_perf_torchrun_cpu():
     assert summary, res.stdout
     s = json.loads(summary[0])
     assert s["world"] == 2 and s["backend"] == "gloo"
+
+
+def test_bench_busbw_ab_companion_cpu():
+    """--ab always exercises the plugin-vs-stock busbw A/B sub-launch
+    machinery (two sub-torchruns, JSON parse, companion embedding) on CPU
+    with gloo — the exact path the driver's N>1 GPU run takes."""
+    res = subprocess.run(
+        [sys.executable, os.path.join(REPO, "bench.py"), "--model",
+         "resnet50", "--steps", "1", "--warmup", "0", "--batch", "2",
+         "--ab", "always", "--ab-iters", "2", "--ab-max-bytes", "4096"],
+        capture_output=True, text=True, timeout=600, cwd=REPO,
+    )
+    assert res.returncode == 0, res.stderr[-2000:]
+    lines = [ln for ln in res.stdout.splitlines() if ln.startswith("{")]
+    assert len(lines) == 1, res.stdout
+    out = json.loads(lines[0])
+    ab = out.get("companion_busbw_ab")
+    assert ab is not None
+    assert ab["ab_status"] == "ok", ab
+    assert ab["sizes"][0] == 8
+    assert len(ab["busbw_plugin_GBps"]) == len(ab["sizes"])
+    assert len(ab["busbw_stock_GBps"]) == len(ab["sizes"])
+    assert all(isinstance(x, (int, float)) for x in ab["busbw_plugin_GBps"])
