@@ -34,6 +34,10 @@ struct Config {
   bool cuda_ptr = true;
   // Pinned staging pool per comm direction, bytes.
   size_t stage_pool = 64ull * 1024 * 1024;
+  // Process-wide cap on pinned staging memory across ALL comms (pools +
+  // dedicated oversize allocations).  Pools shrink (down to 8 MiB) when
+  // the budget runs low instead of pinning GBs on many-comm nodes.
+  size_t pinned_budget = 2ull * 1024 * 1024 * 1024;
   // D2H/H2D pipeline chunk for staging copies.
   uint32_t stage_chunk = 512 * 1024;
   // Staging copy mode: 0 = SDMA (hipMemcpyAsync), 1 = pack kernel.

@@ -36,6 +36,8 @@ const Config& Config::get() {
     c.sockbuf = (int)env_long("BNET_SOCKBUF", c.sockbuf);
     c.cuda_ptr = env_long("BNET_CUDA_PTR", 1) != 0;
     c.stage_pool = (size_t)env_long("BNET_STAGE_POOL", (long)c.stage_pool);
+    c.pinned_budget =
+        (size_t)env_long("BNET_PINNED_BUDGET", (long)c.pinned_budget);
     c.stage_chunk = (uint32_t)env_long("BNET_STAGE_CHUNK", c.stage_chunk);
     if (c.stage_chunk < 65536) c.stage_chunk = 65536;
     c.stage_kernel = (int)env_long("BNET_STAGE_KERNEL", 0);

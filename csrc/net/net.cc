@@ -427,14 +427,14 @@ ncclResult_t Net::isend(void* send_comm, void* data, int size, int tag,
   r->tag = tag;
   r->comm = c;
   if (ptr_type == NCCL_PTR_CUDA && size > 0) {
-    if ((size_t)size > Config::get().stage_pool) {
-      BNET_WARN("isend: %d B exceeds staging pool (%zu B) — raise "
-                "BNET_STAGE_POOL", size, Config::get().stage_pool);
-      return ncclInternalError;  // would retry forever otherwise
-    }
     if (!c->stage_pool) {
-      c->stage_pool = stage_pool_create();
-      if (!c->stage_pool) return ncclInternalError;
+      bool retry = false;
+      c->stage_pool = stage_pool_create(&retry);
+      if (!c->stage_pool) {
+        if (!retry) return ncclInternalError;
+        *request = nullptr;  // pinned budget exhausted — NCCL retries
+        return ncclSuccess;
+      }
     }
     if (!stage_send_begin(c->stage_pool, r, data, (uint32_t)size)) {
       *request = nullptr;  // pool exhausted — retry later
@@ -494,14 +494,14 @@ ncclResult_t Net::irecv(void* recv_comm, int n, void** data, int* sizes,
   r->gpu_done.store(false, std::memory_order_relaxed);
   r->comm = c;
   if (ptr_type == NCCL_PTR_CUDA) {
-    if ((size_t)r->capacity > Config::get().stage_pool) {
-      BNET_WARN("irecv: %u B exceeds staging pool (%zu B) — raise "
-                "BNET_STAGE_POOL", r->capacity, Config::get().stage_pool);
-      return ncclInternalError;
-    }
     if (!c->stage_pool) {
-      c->stage_pool = stage_pool_create();
-      if (!c->stage_pool) return ncclInternalError;
+      bool retry = false;
+      c->stage_pool = stage_pool_create(&retry);
+      if (!c->stage_pool) {
+        if (!retry) return ncclInternalError;
+        *request = nullptr;  // pinned budget exhausted — NCCL retries
+        return ncclSuccess;
+      }
     }
     if (!stage_recv_begin(c->stage_pool, r, r->dst, r->capacity)) {
       *request = nullptr;

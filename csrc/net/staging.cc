@@ -29,9 +29,30 @@ void launch_copy_kernel(void* dst, const void* src, size_t bytes,
       BNET_WARN("%s failed: %s", #call, hipGetErrorString(e_));    \
   } while (0)
 
+// ---- process-wide pinned-memory budget ------------------------------------
+// Per-comm pools and dedicated oversize allocations all draw from one cap
+// (BNET_PINNED_BUDGET): an 8-GPU node with many channels x peers must not
+// pin GBs unboundedly.
+static std::atomic<size_t> g_pinned{0};
+
+static bool pinned_reserve(size_t n) {
+  size_t budget = Config::get().pinned_budget;
+  size_t cur = g_pinned.load(std::memory_order_relaxed);
+  while (true) {
+    if (cur + n > budget) return false;
+    if (g_pinned.compare_exchange_weak(cur, cur + n)) return true;
+  }
+}
+
+static void pinned_release(size_t n) {
+  g_pinned.fetch_sub(n, std::memory_order_relaxed);
+}
+
 struct StageAlloc {
   char* host = nullptr;
   size_t pool_off = 0;
+  bool dedicated = false;  // oversize: own hipHostMalloc, not from the arena
+  size_t ded_cap = 0;      // pinned bytes behind a dedicated allocation
   uint32_t size = 0;
   uint64_t t_begin_ns = 0;   // for slow-event diagnostics
   bool diag_logged = false;
@@ -62,6 +83,50 @@ class StagePool {
   std::vector<hipEvent_t> ev_cache;
   std::vector<StageAlloc*> inflight;  // send stagings with copies pending
   std::atomic<int> pending{0};
+  // one cached dedicated buffer for oversize messages (> pool size), so a
+  // soak of oversize messages does not hipHostMalloc per message
+  char* big_cache = nullptr;
+  size_t big_cache_size = 0;
+  bool warned_oversize = false;
+
+  // Oversize path: take the cached buffer if it fits, else allocate a
+  // dedicated pinned buffer under the global budget.  Returns nullptr if
+  // the budget cannot fit it right now (caller retries); *cap is the
+  // actual pinned allocation size (what free_dedicated must release).
+  char* alloc_dedicated(size_t sz, size_t* cap) {
+    if (big_cache && big_cache_size >= sz) {
+      char* b = big_cache;
+      *cap = big_cache_size;
+      big_cache = nullptr;
+      big_cache_size = 0;
+      return b;
+    }
+    if (!pinned_reserve(sz)) return nullptr;
+    char* b = nullptr;
+    hipError_t e = hipHostMalloc((void**)&b, sz, hipHostMallocDefault);
+    if (e != hipSuccess) {
+      BNET_WARN("oversize hipHostMalloc(%zu) failed: %s", sz,
+                hipGetErrorString(e));
+      pinned_release(sz);
+      return nullptr;
+    }
+    *cap = sz;
+    return b;
+  }
+
+  void free_dedicated(char* b, size_t sz) {
+    if (!big_cache) {  // keep the largest one around for reuse
+      big_cache = b;
+      big_cache_size = sz;
+      return;
+    }
+    if (big_cache_size < sz) {
+      std::swap(big_cache, b);
+      std::swap(big_cache_size, sz);
+    }
+    (void)hipHostFree(b);
+    pinned_release(sz);
+  }
 
   hipEvent_t get_event() {
     if (!ev_cache.empty()) {
@@ -94,12 +159,39 @@ bool staging_available() {
   return avail == 1;
 }
 
-StagePool* stage_pool_create() {
+StagePool* stage_pool_create(bool* retry_later) {
+  if (retry_later) *retry_later = false;
   auto* p = new StagePool();
-  p->size = Config::get().stage_pool;
+  // Budget-clamped pool: shrink (halving, floor 8 MiB) instead of blowing
+  // the process-wide pinned cap when many comms are live.  A pool takes at
+  // most HALF the remaining budget so its peer direction (and later comms)
+  // can still get one — a send pool grabbing the whole budget would
+  // livelock its own recv side behind endless NCCL retries.
+  constexpr size_t kFloor = 8ull * 1024 * 1024;
+  size_t budget = Config::get().pinned_budget;
+  size_t used = g_pinned.load(std::memory_order_relaxed);
+  size_t half_left = budget > used ? (budget - used) / 2 : 0;
+  size_t want =
+      std::min(Config::get().stage_pool, std::max(half_left, kFloor));
+  while (!pinned_reserve(want)) {
+    if (want <= kFloor) {
+      BNET_WARN("pinned budget exhausted (%zu in use of %zu) — staging "
+                "pool creation deferred; closing comms frees budget",
+                g_pinned.load(), Config::get().pinned_budget);
+      delete p;
+      if (retry_later) *retry_later = true;
+      return nullptr;
+    }
+    want /= 2;
+  }
+  if (want < Config::get().stage_pool)
+    BNET_INFO("staging pool clamped to %zu MiB by BNET_PINNED_BUDGET",
+              want >> 20);
+  p->size = want;
   hipError_t e = hipHostMalloc((void**)&p->base, p->size, hipHostMallocDefault);
   if (e != hipSuccess) {
     BNET_WARN("hipHostMalloc(%zu) failed: %s", p->size, hipGetErrorString(e));
+    pinned_release(want);
     delete p;
     return nullptr;
   }
@@ -118,6 +210,11 @@ void stage_pool_destroy(StagePool* p) {
   if (p->d2h) (void)hipStreamDestroy(p->d2h);
   if (p->h2d) (void)hipStreamDestroy(p->h2d);
   if (p->base) (void)hipHostFree(p->base);
+  pinned_release(p->size);
+  if (p->big_cache) {
+    (void)hipHostFree(p->big_cache);
+    pinned_release(p->big_cache_size);
+  }
   delete p;
 }
 
@@ -140,12 +237,34 @@ static bool roctx_on() {
   return on == 1;
 }
 
+// Bounce-space acquisition shared by send/recv begin (pool mutex held):
+// arena for messages that fit the pool, dedicated budget-accounted pinned
+// allocation for oversize ones (no hard failure — the reference-era
+// "exceeds staging pool" ncclInternalError is gone; see staging.h).
+static bool acquire_bounce(StagePool* p, StageAlloc* a, uint32_t need) {
+  if ((size_t)need > p->size) {
+    if (!p->warned_oversize) {
+      p->warned_oversize = true;
+      BNET_WARN("staged message (%u B) exceeds the %zu MiB staging pool — "
+                "using dedicated pinned allocations (raise BNET_STAGE_POOL "
+                "to keep oversize messages in the ring)",
+                need, p->size >> 20);
+    }
+    a->host = p->alloc_dedicated(need, &a->ded_cap);
+    if (!a->host) return false;  // budget full right now — caller retries
+    a->dedicated = true;
+    return true;
+  }
+  a->host = p->alloc(need, &a->pool_off);
+  return a->host != nullptr;
+}
+
 bool stage_send_begin(StagePool* p, SendRequest* req, const void* src,
                       uint32_t total) {
   if (roctx_on()) roctxRangePush("bnet_stage_send_d2h");
   std::lock_guard<std::mutex> lk(p->mu);
   auto* a = new StageAlloc();
-  if (!(a->host = p->alloc(std::max(total, 1u), &a->pool_off))) {
+  if (!acquire_bounce(p, a, std::max(total, 1u))) {
     delete a;
     if (roctx_on()) roctxRangePop();
     return false;
@@ -285,7 +404,7 @@ bool stage_recv_begin(StagePool* p, RecvRequest* req, void* dst,
                       uint32_t capacity) {
   std::lock_guard<std::mutex> lk(p->mu);
   auto* a = new StageAlloc();
-  if (!(a->host = p->alloc(std::max(capacity, 1u), &a->pool_off))) {
+  if (!acquire_bounce(p, a, std::max(capacity, 1u))) {
     delete a;
     return false;
   }
@@ -335,7 +454,10 @@ void stage_release(StagePool* p, SendRequest* req) {
   StageAlloc* a = (StageAlloc*)req->stage;
   if (!a) return;
   std::lock_guard<std::mutex> lk(p->mu);
-  p->free(a->pool_off, a->size);
+  if (a->dedicated)
+    p->free_dedicated(a->host, a->ded_cap);
+  else
+    p->free(a->pool_off, a->size);
   req->stage = nullptr;
   delete a;
 }
@@ -346,9 +468,22 @@ void stage_release(StagePool* p, RecvRequest* req) {
   std::lock_guard<std::mutex> lk(p->mu);
   hipEvent_t ev = a->done_ev.load(std::memory_order_relaxed);
   if (ev) p->put_event(ev);
-  p->free(a->pool_off, a->size);
+  if (a->dedicated)
+    p->free_dedicated(a->host, a->ded_cap);
+  else
+    p->free(a->pool_off, a->size);
   req->stage = nullptr;
   delete a;
+}
+
+// test hooks: process-wide pinned accounting (tests/test_gpu_plugin_cuda.py
+// asserts the budget is respected across comm churn)
+extern "C" {
+__attribute__((visibility("default"))) void bnet_pinned_stats(
+    size_t* used, size_t* budget) {
+  if (used) *used = g_pinned.load(std::memory_order_relaxed);
+  if (budget) *budget = Config::get().pinned_budget;
+}
 }
 
 }  // namespace baguanet

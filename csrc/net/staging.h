@@ -33,7 +33,20 @@ class StagePool;
 bool staging_available();
 
 // ---- pool lifecycle (per comm, lazily created on first CUDA request) ----
-StagePool* stage_pool_create();
+// Pool size is clamped by the process-wide BNET_PINNED_BUDGET (halving to
+// an 8 MiB floor); *retry_later distinguishes a temporarily-exhausted
+// budget (caller should retry: closing comms frees budget) from a hard
+// hipHostMalloc failure.
+//
+// Messages larger than the pool take a DEDICATED budget-accounted pinned
+// allocation (one cached per pool) instead of failing.  Deliberately not a
+// modular windowed ring: RCCL's proxy never posts messages beyond a few
+// MiB (its net messages are bounded by buffSize/NCCL_STEPS slices), so
+// oversize is a configuration-robustness path — a simple contiguous
+// fallback is preferred over wrap-around offset mapping plus sender/
+// receiver window pacing inside the transport hot path, which round-1
+// soak testing showed is the most race-sensitive code in the plugin.
+StagePool* stage_pool_create(bool* retry_later = nullptr);
 void stage_pool_destroy(StagePool* p);
 
 // ---- send path ----
