@@ -57,10 +57,15 @@ def test_rccl_with_plugin_env(tmp_path):
             f.write(out)
     assert res.returncode == 0, f"RCCL run failed:\n{out[-4000:]}"
     assert "RCCL_OK" in out
-    # our v6 symbol must never be reported missing
-    assert "Failed to find ncclNetPlugin_v6" not in out
-    # RCCL resolves librccl-net-bagua.so and must have loaded v6 (net init
-    # runs even at world=1: topology probing enumerates net devices)
+    # none of our exported symbols may be reported missing
+    for v in (6, 7, 8):
+        assert f"Failed to find ncclNetPlugin_v{v}" not in out
+    # RCCL probes v10..v6 newest-first and must pick our v8 vtable (net
+    # init runs even at world=1: topology probing enumerates net devices)
     loaded = [ln for ln in out.splitlines() if "Loaded net plugin" in ln]
     assert loaded, "RCCL did not load the plugin:\n" + out[-3000:]
-    assert any("v6" in ln for ln in loaded), loaded
+    assert any("(v8)" in ln for ln in loaded), loaded
+    # and select it — garbage v8 properties (wrong struct layout) would
+    # break network selection
+    assert any("Using network BaguaNetAMD" in ln
+               for ln in out.splitlines()), out[-3000:]
