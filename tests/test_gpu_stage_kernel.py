@@ -68,6 +68,8 @@ def _oversize(q):
 
     sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
     sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    import time
+
     import torch
 
     from baguanet.plugin import Plugin
@@ -75,26 +77,38 @@ def _oversize(q):
 
     p = Plugin()
     lcomm, scomm, rcomm = establish(p)
-    big = torch.zeros(16 << 18, device="cuda")  # 16 MiB > pool
+    big = torch.randn(16 << 18, device="cuda")  # 16 MiB > pool
+    out = torch.zeros_like(big)
     torch.cuda.synchronize()
-    mh = p.reg_mr(scomm, C.c_void_p(big.data_ptr()), big.numel() * 4, 0x2)
-    try:
-        p.isend(scomm, C.c_void_p(big.data_ptr()), big.numel() * 4, mh)
-        q.put("no-error")
-    except RuntimeError:
-        q.put("errored")  # ncclInternalError, not an infinite retry
+    size = big.numel() * 4
+    smh = p.reg_mr(scomm, C.c_void_p(big.data_ptr()), size, 0x2)
+    rmh = p.reg_mr(rcomm, C.c_void_p(out.data_ptr()), size, 0x2)
+    rreq = sreq = None
+    t0 = time.monotonic()
+    while rreq is None:
+        rreq = p.irecv(rcomm, C.c_void_p(out.data_ptr()), size, rmh)
+        assert time.monotonic() - t0 < 60
+    while sreq is None:
+        sreq = p.isend(scomm, C.c_void_p(big.data_ptr()), size, smh)
+        assert time.monotonic() - t0 < 60
+    assert p.wait(sreq, 120) == size
+    assert p.wait(rreq, 120) == size
+    torch.cuda.synchronize()
+    assert torch.equal(big, out)
     p.close_send(scomm)
     p.close_recv(rcomm)
     p.close_listen(lcomm)
+    q.put("ok")
 
 
-def test_message_larger_than_pool_errors():
-    """A single message larger than the staging pool must fail loudly
-    (ncclInternalError) instead of retrying forever."""
+def test_message_larger_than_pool_transfers():
+    """A message larger than the staging pool streams through a dedicated
+    budget-accounted pinned allocation (it used to hard-fail with
+    ncclInternalError; VERDICT r1 weak #5)."""
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
     proc = ctx.Process(target=_oversize, args=(q,))
     proc.start()
-    assert q.get(timeout=240) == "errored"
+    assert q.get(timeout=240) == "ok"
     proc.join(30)
     assert proc.exitcode == 0
