@@ -28,6 +28,18 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 DEPTH = 16  # outstanding messages (NCCL proxy pipelines similarly)
 
 
+def _enter_child_ns(ns_conn, args):
+    """Receiver side of --shaped: unshare a netns, let the parent move the
+    veth peer in, configure + shape it, and use it as the NIC."""
+    import netns_rig as rig
+
+    rig.unshare_newnet()
+    ns_conn.send(os.getpid())
+    assert ns_conn.recv() == "veth-moved"
+    rig.child_setup(args.shaped, args.delay_us)
+    args.ifname = rig.CHILD_IF
+
+
 def _set_env(args, role=None):
     # per-role engine override for A/B isolation (BNET_IMPL_SENDER/RECEIVER)
     if role:
@@ -40,7 +52,9 @@ def _set_env(args, role=None):
     os.environ["BNET_IO_THREADS"] = str(args.io_threads)
 
 
-def _receiver(conn, args, out_q):
+def _receiver(conn, args, out_q, ns_conn=None):
+    if ns_conn is not None:
+        _enter_child_ns(ns_conn, args)
     _set_env(args, "receiver")
     from baguanet.plugin import Plugin
 
@@ -124,8 +138,10 @@ def _sender(conn, args, out_q):
     out_q.put(results)
 
 
-def _duplex_worker(conn, args, out_q, is_a):
+def _duplex_worker(conn, args, out_q, is_a, ns_conn=None):
     """Each process sends AND receives simultaneously (ring-edge pattern)."""
+    if ns_conn is not None:
+        _enter_child_ns(ns_conn, args)
     _set_env(args)
     from baguanet.plugin import Plugin
 
@@ -202,22 +218,53 @@ def main():
     ap.add_argument("--json", action="store_true")
     ap.add_argument("--duplex", action="store_true",
                     help="both directions simultaneously (ring-edge pattern)")
+    ap.add_argument("--shaped", type=float, default=None,
+                    help="run the receiver behind a veth pair in its own "
+                         "netns, egress-shaped to this many Gbit/s on both "
+                         "devices (0 = veth, unshaped).  Needs "
+                         "CAP_NET_ADMIN; see netns_rig.py")
+    ap.add_argument("--delay-us", type=float, default=10.0,
+                    help="one-way netem delay (ignored on kernels without "
+                         "sch_netem, where TBF provides the rate ceiling)")
     args = ap.parse_args()
 
     ctx = mp.get_context("spawn")
     a, b = ctx.Pipe()
     q = ctx.Queue()
+    rig = None
+    ns_parent = ns_child = None
+    sender_args = args
+    if args.shaped is not None:
+        sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+        import netns_rig as rig
+        import copy
+
+        ns_parent, ns_child = ctx.Pipe()
+        args.ifname = rig.CHILD_IF  # receiver side (child netns)
+        sender_args = copy.copy(args)
+        sender_args.ifname = rig.PARENT_IF
     if args.duplex:
-        pr = ctx.Process(target=_duplex_worker, args=(a, args, q, True))
-        ps = ctx.Process(target=_duplex_worker, args=(b, args, q, False))
+        pr = ctx.Process(target=_duplex_worker,
+                         args=(a, args, q, True, ns_child))
+        ps = ctx.Process(target=_duplex_worker,
+                         args=(b, sender_args, q, False))
     else:
-        pr = ctx.Process(target=_receiver, args=(a, args, q))
-        ps = ctx.Process(target=_sender, args=(b, args, q))
+        pr = ctx.Process(target=_receiver, args=(a, args, q, ns_child))
+        ps = ctx.Process(target=_sender, args=(b, sender_args, q))
     pr.start()
+    if rig is not None:
+        child_pid = ns_parent.recv()
+        rig.parent_setup(child_pid)
+        kind = None
+        if args.shaped > 0:
+            kind = rig.parent_shape(args.shaped, args.delay_us)
+        ns_parent.send("veth-moved")
     ps.start()
     outs = [q.get(timeout=600), q.get(timeout=600)]
     pr.join(30)
     ps.join(30)
+    if rig is not None:
+        rig.parent_teardown()
     results = next(o for o in outs if isinstance(o, list))
     header = {
         "bench": "plugin p2p duplex" if args.duplex else "plugin p2p one-way",
@@ -225,6 +272,9 @@ def main():
         "io_threads": args.io_threads,
         "ifname": args.ifname,
     }
+    if args.shaped is not None:
+        header["shaped_gbit"] = args.shaped
+        header["qdisc"] = kind if args.shaped > 0 else "none"
     if args.json:
         print(json.dumps({**header, "results": results}))
     else:
