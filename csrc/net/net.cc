@@ -460,8 +460,10 @@ ncclResult_t Net::isend(void* send_comm, void* data, int size, int tag,
                               (uint32_t)size);
   int nchunks = r->total ? (int)((r->total + r->chunk - 1) / r->chunk) : 1;
   if (nchunks <= 1) {
+    // single-chunk: statically routed — write it inline from this thread
+    // when the owning socket is idle (post_send), else kick
     size_t nsmall = c->socks.size() < 2 ? c->socks.size() : 2;
-    Engine::get().kick_sock(c->socks[seq % nsmall]);
+    Engine::get().post_send(c->socks[seq % nsmall]);
   }
   else
     Engine::get().kick_comm(c, nchunks);

@@ -219,10 +219,33 @@ void IoThread::set_epollout(TcpSock* s, bool on) {
 
 void IoThread::progress(TcpSock* s) {
   if (s->fd < 0) return;
-  if (s->is_recv)
+  if (s->is_recv) {
     progress_recv(s);
-  else
-    progress_send(s);
+    return;
+  }
+  // owner_busy exclusion against a proxy-thread inline send.  The loser
+  // marks rekick; the winner re-kicks after releasing, so a transition
+  // that raced with the release is never lost.
+  if (s->owner_busy.exchange(1, std::memory_order_acquire) != 0) {
+    s->rekick.store(true, std::memory_order_release);
+    return;
+  }
+  progress_send(s);
+  s->owner_busy.store(0, std::memory_order_release);
+  if (s->rekick.exchange(false, std::memory_order_acq_rel)) kick(s);
+}
+
+bool IoThread::try_inline_send(TcpSock* s) {
+  if (s->fd < 0) return false;
+  if (s->owner_busy.exchange(1, std::memory_order_acquire) != 0)
+    return false;  // owner busy right now — caller falls back to kick()
+  // progress_send is thread-agnostic under owner_busy exclusion: tx state
+  // is exclusively owned, epoll_ctl (set_epollout) is kernel-thread-safe,
+  // kick()/debug counters are atomic or mutex-protected.
+  progress_send(s);
+  s->owner_busy.store(0, std::memory_order_release);
+  if (s->rekick.exchange(false, std::memory_order_acq_rel)) kick(s);
+  return true;
 }
 
 // Claim the next unsent chunk across the comm's active requests, oldest
@@ -637,6 +660,15 @@ void Engine::kick_sock(TcpSock* s) {
   // seq_cst publish-then-recheck pair, stalling a message until the
   // pre-block sweep.  A redundant kick to a busy socket is a cheap
   // no-op; correctness wins.
+  threads_[s->io_thread]->kick(s);
+}
+
+void Engine::post_send(TcpSock* s) {
+  // Proxy-inline first: the posting thread writes the bytes itself when
+  // the socket's owner is idle, skipping the kick->eventfd->IO-thread
+  // wake (~10-20 us).  Contention (or an engine without inline support)
+  // falls back to the unconditional kick.
+  if (threads_[s->io_thread]->try_inline_send(s)) return;
   threads_[s->io_thread]->kick(s);
 }
 

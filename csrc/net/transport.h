@@ -220,6 +220,14 @@ struct TcpSock {
   // the request (seq_cst + fence) then kicks only sockets with snd_idle
   // set — same Dekker publish-then-recheck handshake as recv parking
   std::atomic<bool> snd_idle{false};
+  // proxy-inline-send exclusion: every progress_send entry (IO-thread
+  // event/kick/sweep AND the posting thread's inline attempt from isend)
+  // must win owner_busy first.  The loser marks rekick; the winner re-runs
+  // via kick() after releasing, so no transition is lost.  Removes the
+  // kick->eventfd->IO-thread hop (~10-20 us) from the small-message path —
+  // the same trick as NCCL's own net_socket inline sends.
+  std::atomic<int> owner_busy{0};
+  std::atomic<bool> rekick{false};
   // io_uring engine per-socket state (unused by the epoll engine)
   static constexpr int kUrBatch = 8;  // chunks per WRITEV submission
   struct {
@@ -357,6 +365,11 @@ class IIoThread {
   virtual void add_sock(TcpSock* s) = 0;          // thread-safe
   virtual void remove_sock_sync(TcpSock* s) = 0;  // blocks until removed
   virtual void kick(TcpSock* s) = 0;              // re-run progress for s
+  // Attempt the first send progress inline on the CALLING thread (the
+  // NCCL proxy posting an isend).  Returns false if the engine does not
+  // support it or the socket's owner is busy — caller falls back to
+  // kick().  Only valid for send sockets.
+  virtual bool try_inline_send(TcpSock* s) { (void)s; return false; }
 };
 
 class IoThread : public IIoThread {
@@ -366,6 +379,7 @@ class IoThread : public IIoThread {
   void add_sock(TcpSock* s) override;
   void remove_sock_sync(TcpSock* s) override;
   void kick(TcpSock* s) override;
+  bool try_inline_send(TcpSock* s) override;
 
  private:
   void run();
@@ -406,6 +420,8 @@ class Engine {
   void kick_comm(SendComm* c, int max_socks = -1);
   void kick_comm(RecvComm* c);
   void kick_sock(TcpSock* s);
+  // inline-first send from the posting thread; falls back to kick_sock
+  void post_send(TcpSock* s);
   const char* impl() const { return impl_; }
 
  private:
