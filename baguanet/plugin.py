@@ -247,6 +247,28 @@ class Plugin:
         )
         return req if req.value else None
 
+    def irecv_n(self, rcomm, bufs, sizes, mhandle, tags=None):
+        """Grouped receive: post len(bufs) buffers as ONE request
+        (maxRecvs).  Returns (request, n) — test() fills n sizes."""
+        n = len(bufs)
+        datav = (C.c_void_p * n)(*[C.cast(b, C.c_void_p) for b in bufs])
+        sizev = (C.c_int * n)(*sizes)
+        tagv = (C.c_int * n)(*(tags or [0] * n))
+        mhv = (C.c_void_p * n)(
+            *[mhandle.value if mhandle else None] * n)
+        req = C.c_void_p(None)
+        self._check(
+            self.vt.irecv(rcomm, n, datav, sizev, tagv, mhv, C.byref(req)),
+            "irecv",
+        )
+        return (req if req.value else None), n
+
+    def test_n(self, req, n):
+        done = C.c_int(0)
+        sizes = (C.c_int * n)(*([-1] * n))
+        self._check(self.vt.test(req, C.byref(done), sizes), "test")
+        return bool(done.value), list(sizes)
+
     def iflush(self, rcomm, data, size: int, mhandle):
         datav = (C.c_void_p * 1)(C.cast(data, C.c_void_p))
         sizev = (C.c_int * 1)(size)

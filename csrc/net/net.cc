@@ -70,7 +70,7 @@ Net::Net() {
     d.props.port = 0;
     d.props.latency = 0.0f;
     d.props.maxComms = 65536;  // reference advertised 65536 (nthread:100)
-    d.props.maxRecvs = 1;
+    d.props.maxRecvs = kMaxRecvs;  // grouped receives (n consecutive slots)
     BNET_INFO("baguanet device %s speed %d Mbps pci %s ptrSupport 0x%x",
               d.name_str.c_str(), d.props.speed, d.nif.pci_path.c_str(),
               ptr_support_);
@@ -475,58 +475,82 @@ ncclResult_t Net::irecv(void* recv_comm, int n, void** data, int* sizes,
                         int* tags, void** mhandles, void** request) {
   auto* c = (RecvComm*)recv_comm;
   if (c->error.load(std::memory_order_relaxed)) return ncclSystemError;
-  if (n != 1) {
-    BNET_WARN("irecv: grouped recv n=%d unsupported (maxRecvs=1)", n);
+  if (n < 1 || n > kMaxRecvs) {
+    BNET_WARN("irecv: grouped recv n=%d out of range (maxRecvs=%d)", n,
+              kMaxRecvs);
     return ncclInternalError;
   }
-  RecvRequest* r = &c->reqs[c->post_next % NCCL_NET_MAX_REQUESTS];
-  uint64_t ss0 = r->state_seq.load(std::memory_order_acquire);
-  if (ss_state(ss0) != REQ_FREE) {
-    c->last_refusal_ss = ss0;
-    c->last_refusal_at = c->post_next;
-    *request = nullptr;
-    return ncclSuccess;
-  }
-  int ptr_type = mhandles ? (int)(uintptr_t)mhandles[0] : NCCL_PTR_HOST;
-  r->dst = (char*)data[0];
-  r->capacity = (uint32_t)sizes[0];
-  r->tag = tags ? tags[0] : 0;
-  r->total.store(-1, std::memory_order_relaxed);
-  r->received.store(0, std::memory_order_relaxed);
-  r->gpu_done.store(false, std::memory_order_relaxed);
-  r->comm = c;
-  if (ptr_type == NCCL_PTR_CUDA) {
-    if (!c->stage_pool) {
-      bool retry = false;
-      c->stage_pool = stage_pool_create(&retry);
-      if (!c->stage_pool) {
-        if (!retry) return ncclInternalError;
-        *request = nullptr;  // pinned budget exhausted — NCCL retries
-        return ncclSuccess;
-      }
-    }
-    if (!stage_recv_begin(c->stage_pool, r, r->dst, r->capacity)) {
+  // Grouped receive = n consecutive seq slots; the i-th arriving message
+  // (per-comm FIFO) lands in member i, and the echoed tag verifies the
+  // alignment.  Pre-flight EVERYTHING before publishing member 0: once a
+  // member's state_seq goes ACTIVE, sockets may start landing bytes in it
+  // and rollback is impossible.
+  RecvRequest* members[kMaxRecvs];
+  for (int i = 0; i < n; i++) {
+    RecvRequest* r = &c->reqs[(c->post_next + i) % NCCL_NET_MAX_REQUESTS];
+    uint64_t ss0 = r->state_seq.load(std::memory_order_acquire);
+    if (ss_state(ss0) != REQ_FREE) {
+      c->last_refusal_ss = ss0;
+      c->last_refusal_at = c->post_next + i;
       *request = nullptr;
       return ncclSuccess;
     }
-  } else {
-    r->stage = nullptr;
+    members[i] = r;
   }
-  uint32_t seq = c->post_next;
-  // seq_cst store + seq_cst fence before kick_comm's parked load: pairs
-  // with the parking IO thread's publish-then-recheck (drain_recv) so a
-  // concurrent park cannot be missed
-  r->state_seq.store(pack_ss(seq, REQ_ACTIVE), std::memory_order_seq_cst);
-  std::atomic_thread_fence(std::memory_order_seq_cst);
-  c->post_next++;
-  c->stats.irecv_count.fetch_add(1, std::memory_order_relaxed);
+  bool need_stage = false;
+  for (int i = 0; i < n; i++) {
+    int ptr_type = mhandles ? (int)(uintptr_t)mhandles[i] : NCCL_PTR_HOST;
+    if (ptr_type == NCCL_PTR_CUDA) need_stage = true;
+  }
+  if (need_stage && !c->stage_pool) {
+    bool retry = false;
+    c->stage_pool = stage_pool_create(&retry);
+    if (!c->stage_pool) {
+      if (!retry) return ncclInternalError;
+      *request = nullptr;  // pinned budget exhausted — NCCL retries
+      return ncclSuccess;
+    }
+  }
+  // acquire all staging allocations up front (roll back on failure)
+  for (int i = 0; i < n; i++) {
+    RecvRequest* r = members[i];
+    int ptr_type = mhandles ? (int)(uintptr_t)mhandles[i] : NCCL_PTR_HOST;
+    r->dst = (char*)data[i];
+    r->capacity = (uint32_t)sizes[i];
+    r->tag = tags ? tags[i] : 0;
+    r->total.store(-1, std::memory_order_relaxed);
+    r->received.store(0, std::memory_order_relaxed);
+    r->gpu_done.store(false, std::memory_order_relaxed);
+    r->comm = c;
+    r->group_n = 0;
+    r->stage = nullptr;
+    if (ptr_type == NCCL_PTR_CUDA &&
+        !stage_recv_begin(c->stage_pool, r, r->dst, r->capacity)) {
+      for (int j = 0; j < i; j++)
+        if (members[j]->stage) stage_release(c->stage_pool, members[j]);
+      *request = nullptr;
+      return ncclSuccess;
+    }
+  }
+  members[0]->group_n = (uint8_t)n;
   auto& T = Telemetry::get();
-  T.irecv_count.fetch_add(1, std::memory_order_relaxed);
-  T.hist_add(T.irecv_hist, (uint64_t)sizes[0]);
-  r->span_slot = T.span_begin(1, (uint64_t)(uintptr_t)c, seq,
-                              (uint32_t)sizes[0]);
-  Engine::get().kick_comm(c);  // wake sockets parked on this seq
-  *request = tag_recv(r);
+  for (int i = 0; i < n; i++) {
+    RecvRequest* r = members[i];
+    uint32_t seq = c->post_next + i;
+    // seq_cst store + seq_cst fence before kick_comm's parked load: pairs
+    // with the parking IO thread's publish-then-recheck (drain_recv) so a
+    // concurrent park cannot be missed
+    r->state_seq.store(pack_ss(seq, REQ_ACTIVE), std::memory_order_seq_cst);
+    T.hist_add(T.irecv_hist, (uint64_t)sizes[i]);
+    r->span_slot = T.span_begin(1, (uint64_t)(uintptr_t)c, seq,
+                                (uint32_t)sizes[i]);
+  }
+  std::atomic_thread_fence(std::memory_order_seq_cst);
+  c->post_next += (uint32_t)n;
+  c->stats.irecv_count.fetch_add(n, std::memory_order_relaxed);
+  T.irecv_count.fetch_add(n, std::memory_order_relaxed);
+  Engine::get().kick_comm(c);  // wake sockets parked on these seqs
+  *request = tag_recv(members[0]);
   return ncclSuccess;
 }
 
@@ -572,16 +596,28 @@ ncclResult_t Net::test(void* request, int* done, int* sizes) {
       BNET_WARN("test(recv): comm error %d", c->error.load());
       return ncclSystemError;
     }
-    if (r->socket_complete() && (!r->stage || stage_recv_done(r))) {
-      *done = 1;
-      if (sizes) sizes[0] = (int)r->total.load(std::memory_order_acquire);
-      auto& T = Telemetry::get();
-      T.bytes_recv.fetch_add((uint64_t)r->total.load(),
+    // grouped receive completes when ALL members do (n consecutive slots)
+    int n = r->group_n ? r->group_n : 1;
+    uint32_t s0 = ss_seq(r->state_seq.load(std::memory_order_relaxed));
+    RecvRequest* members[kMaxRecvs];
+    for (int i = 0; i < n; i++) {
+      RecvRequest* m = &c->reqs[(s0 + i) % NCCL_NET_MAX_REQUESTS];
+      if (!m->socket_complete() || (m->stage && !stage_recv_done(m)))
+        return ncclSuccess;  // not done yet
+      members[i] = m;
+    }
+    *done = 1;
+    auto& T = Telemetry::get();
+    for (int i = 0; i < n; i++) {
+      RecvRequest* m = members[i];
+      if (sizes) sizes[i] = (int)m->total.load(std::memory_order_acquire);
+      T.bytes_recv.fetch_add((uint64_t)m->total.load(),
                              std::memory_order_relaxed);
-      T.span_end(r->span_slot);
-      if (r->stage) stage_release(c->stage_pool, r);
-      uint64_t ss = r->state_seq.load(std::memory_order_relaxed);
-      r->state_seq.store(pack_ss(ss_seq(ss), REQ_FREE),
+      T.span_end(m->span_slot);
+      if (m->stage) stage_release(c->stage_pool, m);
+      m->group_n = 0;
+      uint64_t ss = m->state_seq.load(std::memory_order_relaxed);
+      m->state_seq.store(pack_ss(ss_seq(ss), REQ_FREE),
                          std::memory_order_release);
     }
     return ncclSuccess;

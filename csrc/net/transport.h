@@ -93,6 +93,11 @@ static_assert(sizeof(ListenHandle) <= NCCL_NET_HANDLE_MAXSIZE, "handle size");
 
 enum ReqState : uint32_t { REQ_FREE = 0, REQ_ACTIVE = 1 };
 
+// Grouped-receive ceiling advertised as properties.maxRecvs (the proxy
+// aggregates small recvs when this is > 1; NCCL's IB plugin uses 8).
+// Each group member consumes one of the NCCL_NET_MAX_REQUESTS seq slots.
+constexpr int kMaxRecvs = 4;
+
 // state+seq packed into one atomic word so observers can never see a torn
 // (state, seq) pair.  Without this, a claim scan could read stale
 // state=FREE together with a freshly-written new seq during slot re-posting
@@ -155,6 +160,10 @@ struct RecvRequest {
   char* dst = nullptr;     // user destination (host) — staging writes here
   uint32_t capacity = 0;   // posted buffer size (recv may be smaller)
   int tag = 0;             // posted NCCL tag; checked against ChunkHdr.tag
+  // Grouped receive (irecv n>1, maxRecvs): the FIRST member records the
+  // group size; members occupy n consecutive seq slots and the returned
+  // request completes when all members do.  0/1 = ungrouped.
+  uint8_t group_n = 0;
   std::atomic<int64_t> total{-1};      // from first chunk header
   std::atomic<uint32_t> received{0};   // socket bytes landed
   std::atomic<bool> gpu_done{false};   // H2D staging drained (CUDA dst)

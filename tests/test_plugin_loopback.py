@@ -46,7 +46,7 @@ def test_init_devices_properties(plugin):
     assert props["name"] == "lo"
     assert props["ptrSupport"] & 0x1  # NCCL_PTR_HOST
     assert props["maxComms"] > 0
-    assert props["maxRecvs"] == 1
+    assert props["maxRecvs"] == 4  # grouped receives
     assert props["speed"] > 0
 
 
