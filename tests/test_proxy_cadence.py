@@ -153,6 +153,9 @@ def _abandoned_connect_reaped(env, q):
     def nfds():
         return len(os.listdir("/proc/self/fd"))
 
+    import socket
+    import struct
+
     p = Plugin(abi=8)
     # warm-up cycle so the engine's lazily-created IO threads (epoll +
     # eventfd per thread) are in the baseline fd count
@@ -162,16 +165,32 @@ def _abandoned_connect_reaped(env, q):
     p.close_recv(wr)
     p.close_listen(wl)
     time.sleep(0.1)
-    base = nfds()
 
+    # A loopback connect completes instantly, so to model RCCL abandoning
+    # an IN-PROGRESS establishment the listener's accept queue must be
+    # full (BNET_BACKLOG=1 + prefilled raw connections): further SYNs are
+    # dropped and the plugin's connect() stays pending.
     handle, lcomm = p.listen(0)
-    # start a connect (stashes a task with nstreams sockets), then abandon
-    p.connect(0, handle)
+    magic, family, port_be = struct.unpack_from("<IHH", bytes(handle))
+    port = socket.ntohs(port_be)
+    prefill = []
+    for _ in range(3):
+        rs = socket.socket()
+        rs.setblocking(False)
+        try:
+            rs.connect(("127.0.0.1", port))
+        except BlockingIOError:
+            pass
+        prefill.append(rs)
+    time.sleep(0.2)
+
+    base = nfds()
+    assert p.connect(0, handle) is None  # stays in progress (queue full)
     assert nfds() > base + 1
     time.sleep(0.4)  # > abandon window
     # an unrelated connect on a fresh handle triggers the reap
     handle2, lcomm2 = p.listen(0)
-    scomm, rcomm = _pump(p, handle2, lcomm2)
+    scomm, rcomm = _pump(p, handle2, lcomm2, timeout=60)
     _xfer_ok(p, scomm, rcomm)
     p.close_send(scomm)
     p.close_recv(rcomm)
@@ -179,7 +198,10 @@ def _abandoned_connect_reaped(env, q):
     p.close_listen(lcomm)
     time.sleep(0.1)
     # the abandoned task's sockets must be gone
-    assert nfds() <= base + 1, f"leaked fds: {nfds()} vs base {base}"
+    got = nfds()
+    for rs in prefill:
+        rs.close()
+    assert got <= base + 1, f"leaked fds: {got} vs base {base}"
     q.put("ok")
 
 
@@ -187,4 +209,5 @@ def test_abandoned_connect_reaped():
     assert _run_sub(_abandoned_connect_reaped, {
         "NCCL_SOCKET_IFNAME": "lo",
         "BNET_CONNECT_ABANDON_MS": "200",
+        "BNET_BACKLOG": "1",
     }) == "ok"
