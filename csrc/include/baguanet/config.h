@@ -63,6 +63,10 @@ struct Config {
   std::string implement = "EPOLL";
   // Metrics dump file ("" = disabled); written at process exit.
   std::string metrics_file;
+  // Live Prometheus pull endpoint on 127.0.0.1:(port + rank); 0 = off.
+  // (The reference pushed to a gateway; this image has no egress, so the
+  // live equivalent is scrape-style pull.)
+  int metrics_port = 0;
   // Chrome-trace span dump file ("" = disabled).
   std::string trace_file;
 

@@ -50,6 +50,7 @@ const Config& Config::get() {
     c.implement = env_str("BNET_IMPLEMENT", "EPOLL");
     c.rank = (int)env_long("BNET_RANK", env_long("RANK", -1));
     c.metrics_file = env_str("BNET_METRICS_FILE", "");
+    c.metrics_port = (int)env_long("BNET_METRICS_PORT", 0);
     c.trace_file = env_str("BNET_TRACE_FILE", "");
     // Rank-template the dump paths so N-rank jobs leave N distinguishable
     // files instead of overwriting one another: "%r" substitutes the rank;

@@ -1,11 +1,20 @@
 #include "telemetry.h"
 
+#include <arpa/inet.h>
+#include <errno.h>
+#include <netinet/in.h>
+#include <netinet/tcp.h>
+#include <string.h>
+#include <sys/socket.h>
 #include <time.h>
+#include <unistd.h>
 
 #include <cstdio>
 #include <cstdlib>
+#include <thread>
 
 #include "baguanet/config.h"
+#include "baguanet/log.h"
 #include "transport.h"
 
 namespace baguanet {
@@ -16,6 +25,55 @@ uint64_t now_ns() {
   timespec ts;
   clock_gettime(CLOCK_MONOTONIC, &ts);
   return (uint64_t)ts.tv_sec * 1000000000ull + ts.tv_nsec;
+}
+
+// Live pull endpoint: a detached thread serving the Prometheus text
+// rendering over HTTP on 127.0.0.1:BNET_METRICS_PORT (the reference
+// PUSHED to a gateway every 200 us, nthread:183-211; this image has no
+// egress, so scrape-style pull is the live equivalent).  Multi-rank jobs
+// use port+rank so every rank stays scrapable.
+static void metrics_server(int port) {
+  int fd = socket(AF_INET, SOCK_STREAM | SOCK_CLOEXEC, 0);
+  if (fd < 0) return;
+  int one = 1;
+  setsockopt(fd, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
+  sockaddr_in a{};
+  a.sin_family = AF_INET;
+  a.sin_port = htons((uint16_t)port);
+  a.sin_addr.s_addr = htonl(INADDR_LOOPBACK);
+  if (bind(fd, (sockaddr*)&a, sizeof(a)) < 0 || listen(fd, 16) < 0) {
+    BNET_WARN("metrics endpoint bind(127.0.0.1:%d) failed: %s", port,
+              strerror(errno));
+    close(fd);
+    return;
+  }
+  BNET_INFO("metrics endpoint live on 127.0.0.1:%d", port);
+  while (true) {
+    int c = accept(fd, nullptr, nullptr);
+    if (c < 0) {
+      if (errno == EINTR) continue;
+      break;
+    }
+    char req[1024];
+    (void)!read(c, req, sizeof(req));  // drain the request line
+    char* body = nullptr;
+    size_t blen = 0;
+    FILE* mem = open_memstream(&body, &blen);
+    if (mem) {
+      Telemetry::get().render_metrics(mem);
+      fclose(mem);
+      char hdr[160];
+      int hn = snprintf(hdr, sizeof(hdr),
+                        "HTTP/1.0 200 OK\r\nContent-Type: text/plain; "
+                        "version=0.0.4\r\nContent-Length: %zu\r\n\r\n",
+                        blen);
+      (void)!write(c, hdr, hn);
+      (void)!write(c, body, blen);
+      free(body);
+    }
+    close(c);
+  }
+  close(fd);
 }
 
 Telemetry& Telemetry::get() {
@@ -29,6 +87,11 @@ Telemetry& Telemetry::get() {
       if (!c.trace_file.empty())
         Telemetry::get().dump_trace(c.trace_file.c_str());
     });
+    if (Config::get().metrics_port > 0) {
+      int port = Config::get().metrics_port +
+                 (Config::get().rank > 0 ? Config::get().rank : 0);
+      std::thread(metrics_server, port).detach();
+    }
     return p;
   }();
   return *t;
@@ -63,6 +126,11 @@ void Telemetry::span_end(uint32_t slot) {
 void Telemetry::dump_metrics(const char* path) {
   FILE* f = fopen(path, "w");
   if (!f) return;
+  render_metrics(f);
+  fclose(f);
+}
+
+void Telemetry::render_metrics(FILE* f) {
   // rank label (reference pushed Prometheus with a `rank` label,
   // nthread:183-211) — empty when the launcher exported no RANK
   char rl[32] = "";
@@ -100,7 +168,6 @@ void Telemetry::dump_metrics(const char* path) {
     fprintf(f, "%s_bucket{%sle=\"+Inf\"} %llu\n", names[h], rli,
             (unsigned long long)cum);
   }
-  fclose(f);
 }
 
 void Telemetry::dump_trace(const char* path) {

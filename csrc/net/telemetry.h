@@ -14,6 +14,7 @@
 
 #include <atomic>
 #include <cstdint>
+#include <cstdio>
 
 namespace baguanet {
 
@@ -50,6 +51,7 @@ struct Telemetry {
                       uint32_t nbytes);
   void span_end(uint32_t slot);
   void dump_metrics(const char* path);
+  void render_metrics(FILE* f);  // Prometheus text (file or live endpoint)
   void dump_trace(const char* path);
 };
 

@@ -125,3 +125,77 @@ def test_atexit_file_dumps(tmp_path):
     spans = json.loads(tfile.read_text())
     events = spans["traceEvents"] if isinstance(spans, dict) else spans
     assert any(str(ev.get("name", "")).startswith("isend") for ev in events)
+
+
+def _live_endpoint(env, q):
+    for k, v in env.items():
+        os.environ[k] = v
+    import ctypes as C
+    import time
+    import urllib.request
+
+    from baguanet.plugin import Plugin
+
+    p = Plugin()
+    # one quick loopback message so counters are non-zero
+    handle, lcomm = p.listen(0)
+    scomm = rcomm = None
+    t0 = time.monotonic()
+    while scomm is None or rcomm is None:
+        assert time.monotonic() - t0 < 30
+        scomm = scomm or p.connect(0, handle)
+        rcomm = rcomm or p.accept(lcomm)
+    buf = C.create_string_buffer(b"live-metrics", 12)
+    rbuf = C.create_string_buffer(12)
+    mh = p.reg_mr(scomm, None, 0)
+    rreq = p.irecv(rcomm, rbuf, 12, mh)
+    sreq = p.isend(scomm, buf, 12, mh)
+    p.wait(sreq, 30)
+    p.wait(rreq, 30)
+    port = int(env["BNET_METRICS_PORT"]) + int(env.get("RANK", 0))
+    text = None
+    for _ in range(50):
+        try:
+            text = urllib.request.urlopen(
+                f"http://127.0.0.1:{port}/metrics", timeout=2
+            ).read().decode()
+            break
+        except OSError:
+            time.sleep(0.1)
+    assert text is not None, "metrics endpoint never came up"
+    assert 'bnet_isend_total{rank="2"} 1' in text, text[:500]
+    # scrape again: counters are live, not a one-shot snapshot
+    sreq = p.isend(scomm, buf, 12, mh)
+    rreq = p.irecv(rcomm, rbuf, 12, mh)
+    p.wait(sreq, 30)
+    p.wait(rreq, 30)
+    text2 = urllib.request.urlopen(
+        f"http://127.0.0.1:{port}/metrics", timeout=2).read().decode()
+    assert 'bnet_isend_total{rank="2"} 2' in text2, text2[:500]
+    p.close_send(scomm)
+    p.close_recv(rcomm)
+    p.close_listen(lcomm)
+    q.put("ok")
+
+
+def test_live_metrics_endpoint():
+    """BNET_METRICS_PORT serves live Prometheus text on 127.0.0.1:(port+rank)
+    — the pull-style equivalent of the reference's push-gateway uploader."""
+    import multiprocessing as mp
+    import socket
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    proc = ctx.Process(target=_live_endpoint, args=({
+        "NCCL_SOCKET_IFNAME": "lo",
+        "BNET_METRICS_PORT": str(port - 2),
+        "RANK": "2",
+    }, q))
+    proc.start()
+    assert q.get(timeout=120) == "ok"
+    proc.join(30)
+    assert proc.exitcode == 0
