@@ -97,3 +97,41 @@ def test_gpu_mixed_host_and_cuda(gconn):
     gpu_xfer(plugin, scomm, rcomm, src, dst)
     torch.cuda.synchronize()
     assert torch.equal(src, dst)
+
+
+def test_gpu_grouped_recv_staged(gconn):
+    """Grouped irecv (maxRecvs) with NCCL_PTR_CUDA buffers: three staged
+    GPU messages land in one grouped request."""
+    plugin, scomm, rcomm = gconn
+    torch.cuda.synchronize()
+    srcs = [torch.randn(5000 + 999 * i, device="cuda") for i in range(3)]
+    dsts = [torch.zeros_like(s) for s in srcs]
+    sizes = [s.numel() * 4 for s in srcs]
+    torch.cuda.synchronize()
+    smh = plugin.reg_mr(scomm, C.c_void_p(srcs[0].data_ptr()), sizes[0],
+                        NCCL_PTR_CUDA)
+    rmh = plugin.reg_mr(rcomm, C.c_void_p(dsts[0].data_ptr()), sizes[0],
+                        NCCL_PTR_CUDA)
+    req = None
+    t0 = time.monotonic()
+    while req is None:
+        req, n = plugin.irecv_n(
+            rcomm, [C.c_void_p(d.data_ptr()) for d in dsts], sizes, rmh,
+            tags=[1, 2, 3])
+        assert time.monotonic() - t0 < 30
+    for s, size, tag in zip(srcs, sizes, (1, 2, 3)):
+        sreq = None
+        while sreq is None:
+            sreq = plugin.isend(scomm, C.c_void_p(s.data_ptr()), size, smh,
+                                tag=tag)
+            assert time.monotonic() - t0 < 60
+        assert plugin.wait(sreq, 60) == size
+    while True:
+        done, got = plugin.test_n(req, 3)
+        if done:
+            break
+        assert time.monotonic() - t0 < 60
+    assert got == sizes
+    torch.cuda.synchronize()
+    for i, (s, d) in enumerate(zip(srcs, dsts)):
+        assert torch.equal(s, d), f"grouped member {i} corrupted"
