@@ -63,6 +63,7 @@ def main():
                 total += 1 << 20
             dt = time.perf_counter() - t0
             res["dmabuf_cpu_read_GBps"] = round(total / dt / 1e9, 3)
+            view.release()
             m.close()
         except (OSError, ValueError) as e:
             res["dmabuf_mmap"] = f"failed: {e}"
