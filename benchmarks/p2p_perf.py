@@ -52,6 +52,14 @@ def _set_env(args, role=None):
     os.environ["BNET_IO_THREADS"] = str(args.io_threads)
 
 
+def _pattern(i: int, size: int):
+    """Deterministic per-message payload (numpy-vectorized for soak rates)."""
+    import numpy as np
+
+    return ((np.arange(size, dtype=np.uint32) * 31 + i * 131) & 0xFF).astype(
+        np.uint8).tobytes()
+
+
 def _receiver(conn, args, out_q, ns_conn=None):
     if ns_conn is not None:
         _enter_child_ns(ns_conn, args)
@@ -65,30 +73,51 @@ def _receiver(conn, args, out_q, ns_conn=None):
     while rcomm is None:
         rcomm = p.accept(lcomm)
     mh = p.reg_mr(rcomm, None, 0)
+    verified = 0
     for size in args.sizes:
         n_msgs = max(4, min(args.max_msgs, args.bytes_per_size // size))
         bufs = [C.create_string_buffer(size) for _ in range(DEPTH)]
         conn.send(("ready", size))  # buffers allocated — sender may start
         done = 0
         posted = 0
-        reqs = []
+        reqs = []  # (req, msg_idx)
+        # completion can be out of order across the window, so a slot is
+        # reusable only when ITS request finished — not when any DEPTH
+        # requests have (verify mode checks content per slot)
+        slot_free = [True] * DEPTH
         while done < n_msgs:
-            while posted < n_msgs and len(reqs) < DEPTH:
+            while posted < n_msgs and len(reqs) < DEPTH and \
+                    slot_free[posted % DEPTH]:
                 r = p.irecv(rcomm, bufs[posted % DEPTH], size, mh)
                 if r is None:
                     break
-                reqs.append(r)
+                slot_free[posted % DEPTH] = False
+                reqs.append((r, posted))
                 posted += 1
-            for r in list(reqs):
-                ok, _ = p.test(r)
+            for item in list(reqs):
+                ok, _ = p.test(item[0])
                 if ok:
-                    reqs.remove(r)
+                    if args.verify:
+                        i = item[1]
+                        want = _pattern(i, size)
+                        got = bufs[i % DEPTH].raw[:size]
+                        if got != want:
+                            match = [j for j in range(n_msgs)
+                                     if _pattern(j, size) == got]
+                            raise AssertionError(
+                                f"payload corrupted: size={size} msg={i} "
+                                f"slot={i % DEPTH} content-matches-msgs="
+                                f"{match[:5]} outstanding="
+                                f"{[x[1] for x in reqs]}")
+                        verified += 1
+                    slot_free[item[1] % DEPTH] = True
+                    reqs.remove(item)
                     done += 1
         conn.send(("size-done", size))
     conn.recv()
     p.close_recv(rcomm)
     p.close_listen(lcomm)
-    out_q.put("recv-ok")
+    out_q.put(f"recv-ok verified={verified}" if args.verify else "recv-ok")
 
 
 def _sender(conn, args, out_q):
@@ -105,24 +134,39 @@ def _sender(conn, args, out_q):
     results = []
     for size in args.sizes:
         n_msgs = max(4, min(args.max_msgs, args.bytes_per_size // size))
-        buf = C.create_string_buffer(os.urandom(size), size)
+        if args.verify:
+            bufs = [C.create_string_buffer(size) for _ in range(DEPTH)]
+        else:
+            buf = C.create_string_buffer(os.urandom(size), size)
         tag, s = conn.recv()
         assert tag == "ready" and s == size
         t0 = time.perf_counter()
         done = 0
         posted = 0
-        reqs = []
+        reqs = []  # (req, msg_idx)
+        slot_free = [True] * DEPTH  # see receiver: per-slot gating
         while done < n_msgs:
             while posted < n_msgs and len(reqs) < DEPTH:
-                r = p.isend(scomm, buf, size, mh)
+                if args.verify:
+                    if not slot_free[posted % DEPTH]:
+                        break
+                    sb = bufs[posted % DEPTH]
+                    sb.raw = _pattern(posted, size)
+                    r = p.isend(scomm, sb, size, mh)
+                else:
+                    r = p.isend(scomm, buf, size, mh)
                 if r is None:
                     break
-                reqs.append(r)
+                if args.verify:
+                    slot_free[posted % DEPTH] = False
+                reqs.append((r, posted))
                 posted += 1
-            for r in list(reqs):
-                ok, _ = p.test(r)
+            for item in list(reqs):
+                ok, _ = p.test(item[0])
                 if ok:
-                    reqs.remove(r)
+                    if args.verify:
+                        slot_free[item[1] % DEPTH] = True
+                    reqs.remove(item)
                     done += 1
         # wait for receiver to fully drain this size
         tag, s = conn.recv()
@@ -226,7 +270,12 @@ def main():
     ap.add_argument("--delay-us", type=float, default=10.0,
                     help="one-way netem delay (ignored on kernels without "
                          "sch_netem, where TBF provides the rate ceiling)")
+    ap.add_argument("--verify", action="store_true",
+                    help="content-verify every message (deterministic "
+                         "per-message patterns; soak mode, not peak rate)")
     args = ap.parse_args()
+    if args.verify and args.duplex:
+        ap.error("--verify supports the one-way mode")
 
     ctx = mp.get_context("spawn")
     a, b = ctx.Pipe()
