@@ -170,3 +170,59 @@ def test_unused_parameter_raises():
     m(torch.randn(2, 4)).sum().backward()
     with pytest.raises(RuntimeError, match="received no gradient"):
         m.finish_backward()
+
+
+def test_bucket_ddp_world4_gloo(tmp_path):
+    """World-4 data-parallel numerics (the driver's 8-GPU shape, scaled
+    down): BucketedDDP gradients equal the single-process reference."""
+    import subprocess
+    import sys
+
+    REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+    code = r'''
+import os, torch, torch.distributed as dist, sys
+sys.path.insert(0, os.environ["BNET_REPO"])
+from baguanet.models import resnet50
+from baguanet.parallel import BucketedDDP
+
+dist.init_process_group("gloo")
+rank, world = dist.get_rank(), dist.get_world_size()
+torch.manual_seed(1234)
+model = resnet50(num_classes=10)
+ref = resnet50(num_classes=10)
+ref.load_state_dict(model.state_dict())
+ddp = BucketedDDP(model, bucket_cap_mb=5.0)
+
+torch.manual_seed(99)
+xs = [torch.randn(2, 3, 64, 64) for _ in range(world)]
+ys = [torch.randint(0, 10, (2,)) for _ in range(world)]
+
+ddp.zero_grad()
+loss = torch.nn.functional.cross_entropy(ddp(xs[rank]), ys[rank])
+loss.backward()
+ddp.finish_backward()
+
+# single-process reference: mean gradient over all shards
+ref.zero_grad()
+for x, y in zip(xs, ys):
+    torch.nn.functional.cross_entropy(ref(x), y).div(world).backward()
+
+for (n, p), (_, rp) in zip(ddp.module.named_parameters(),
+                           ref.named_parameters()):
+    assert torch.allclose(p.grad, rp.grad, atol=2e-5), n
+if rank == 0:
+    print("WORLD4_OK")
+'''
+    script = tmp_path / "world4_ddp.py"
+    script.write_text(code)
+    env = dict(os.environ)
+    env["BNET_REPO"] = REPO
+    res = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "4", "--master-addr", "127.0.0.1",
+         "--master-port", "29571", str(script)],
+        capture_output=True, text=True, timeout=600, env=env,
+    )
+    assert res.returncode == 0, res.stderr[-3000:]
+    assert "WORLD4_OK" in res.stdout
