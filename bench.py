@@ -76,7 +76,7 @@ def run_busbw_ab(world: int, iters: int, use_cuda: bool,
         env = _clean_env()  # allreduce_perf auto-selects gloo without CUDA
         try:
             p = subprocess.run(cmd, env=env, capture_output=True, text=True,
-                               timeout=600)
+                               timeout=300)
             if p.returncode != 0:
                 out["ab_status"] = (
                     f"{name} failed rc={p.returncode}: "
@@ -233,6 +233,13 @@ def main():
     want_ab = args.ab == "always" or (args.ab == "auto" and world > 1
                                       and use_cuda)
     if want_ab:
+        # forensic backup on stderr first: if an outer watchdog kills the
+        # process during the A/B sweep, the timed result is not lost
+        print("BENCH_PRELIMINARY "
+              + json.dumps({"value": round(total_img_per_sec, 1),
+                            "ms_per_step": round(ms_per_step, 2),
+                            "n_gpus": n_gpus}),
+              file=sys.stderr, flush=True)
         if world > 1:
             time.sleep(5.0)  # let sibling ranks exit and release their GPUs
         companion = run_busbw_ab(max(world, 1), args.ab_iters, use_cuda,
