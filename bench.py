@@ -123,6 +123,10 @@ def main():
                     default=True,
                     help="NHWC memory format (MIOpen igemm path; +9%% on "
                          "VGG16 fp32, same numerics)")
+    ap.add_argument("--find-mode", action="store_true",
+                    help="torch.backends.cudnn.benchmark=True (MIOpen "
+                         "exhaustive find; slower warmup, possibly faster "
+                         "convs)")
     ap.add_argument("--ab", choices=["auto", "always", "never"],
                     default="auto",
                     help="plugin-vs-stock all_reduce_perf busbw companion "
@@ -154,6 +158,8 @@ def main():
     n_gpus = world if world > 1 else args.gpus
 
     use_cuda = torch.cuda.is_available()
+    if args.find_mode:
+        torch.backends.cudnn.benchmark = True
     device = torch.device(f"cuda:{local_rank}" if use_cuda else "cpu")
     if use_cuda:
         torch.cuda.set_device(device)
