@@ -182,6 +182,114 @@ def _sender(conn, args, out_q):
     out_q.put(results)
 
 
+def _receiver_pairs(conn, args, out_q, ns_conn=None):
+    """--pairs N: N independent comms between the same two processes, all
+    receiving concurrently (models N channels/peers sharing one NIC — the
+    fairness regime the reference optimized for)."""
+    if ns_conn is not None:
+        _enter_child_ns(ns_conn, args)
+    _set_env(args, "receiver")
+    from baguanet.plugin import Plugin
+
+    p = Plugin()
+    handle, lcomm = p.listen(0)
+    conn.send(bytes(handle))
+    rcomms = []
+    while len(rcomms) < args.pairs:
+        rc = p.accept(lcomm)
+        if rc is not None:
+            rcomms.append(rc)
+    mh = p.reg_mr(rcomms[0], None, 0)
+    size = args.sizes[0]
+    n_msgs = max(4, min(args.max_msgs, args.bytes_per_size // size))
+    bufs = [[C.create_string_buffer(size) for _ in range(DEPTH)]
+            for _ in rcomms]
+    conn.send(("ready", size))
+    done = [0] * args.pairs
+    posted = [0] * args.pairs
+    reqs = [[] for _ in rcomms]
+    while any(d < n_msgs for d in done):
+        for ci, rc in enumerate(rcomms):
+            while posted[ci] < n_msgs and len(reqs[ci]) < DEPTH:
+                r = p.irecv(rc, bufs[ci][posted[ci] % DEPTH], size, mh)
+                if r is None:
+                    break
+                reqs[ci].append(r)
+                posted[ci] += 1
+            for r in list(reqs[ci]):
+                ok, _ = p.test(r)
+                if ok:
+                    reqs[ci].remove(r)
+                    done[ci] += 1
+    conn.send(("size-done", size))
+    conn.recv()
+    for rc in rcomms:
+        p.close_recv(rc)
+    p.close_listen(lcomm)
+    out_q.put("recv-ok")
+
+
+def _sender_pairs(conn, args, out_q):
+    _set_env(args, "sender")
+    from baguanet.plugin import Plugin
+
+    p = Plugin()
+    hb = conn.recv()
+    handle_t = C.c_char * len(hb)
+    scomms = []
+    handles = [handle_t.from_buffer_copy(hb) for _ in range(args.pairs)]
+    pending = list(range(args.pairs))
+    while pending:
+        for i in list(pending):
+            sc = p.connect(0, handles[i])
+            if sc is not None:
+                scomms.append(sc)
+                pending.remove(i)
+    mh = p.reg_mr(scomms[0], None, 0)
+    size = args.sizes[0]
+    n_msgs = max(4, min(args.max_msgs, args.bytes_per_size // size))
+    buf = C.create_string_buffer(os.urandom(size), size)
+    tag, s = conn.recv()
+    assert tag == "ready" and s == size
+    t0 = time.perf_counter()
+    done = [0] * args.pairs
+    posted = [0] * args.pairs
+    reqs = [[] for _ in scomms]
+    t_done = [None] * args.pairs
+    while any(d < n_msgs for d in done):
+        for ci, sc in enumerate(scomms):
+            while posted[ci] < n_msgs and len(reqs[ci]) < DEPTH:
+                r = p.isend(sc, buf, size, mh)
+                if r is None:
+                    break
+                reqs[ci].append(r)
+                posted[ci] += 1
+            for r in list(reqs[ci]):
+                ok, _ = p.test(r)
+                if ok:
+                    reqs[ci].remove(r)
+                    done[ci] += 1
+                    if done[ci] == n_msgs:
+                        t_done[ci] = time.perf_counter() - t0
+    tag, s = conn.recv()
+    assert tag == "size-done" and s == size
+    total_dt = time.perf_counter() - t0
+    per_comm = [round(n_msgs * size / dt / 1e9, 3) for dt in t_done]
+    results = [{
+        "size": size, "msgs_per_comm": n_msgs, "pairs": args.pairs,
+        "secs": round(total_dt, 4),
+        "GBps_aggregate": round(args.pairs * n_msgs * size / total_dt / 1e9,
+                                3),
+        "GBps_per_comm": per_comm,
+        # min/max completion-rate ratio: 1.0 = perfectly fair service
+        "fairness": round(min(per_comm) / max(per_comm), 3),
+    }]
+    conn.send("done")
+    for sc in scomms:
+        p.close_send(sc)
+    out_q.put(results)
+
+
 def _duplex_worker(conn, args, out_q, is_a, ns_conn=None):
     """Each process sends AND receives simultaneously (ring-edge pattern)."""
     if ns_conn is not None:
@@ -273,9 +381,12 @@ def main():
     ap.add_argument("--verify", action="store_true",
                     help="content-verify every message (deterministic "
                          "per-message patterns; soak mode, not peak rate)")
+    ap.add_argument("--pairs", type=int, default=1,
+                    help="N concurrent comms between the two processes "
+                         "(fairness measurement; uses the FIRST size only)")
     args = ap.parse_args()
-    if args.verify and args.duplex:
-        ap.error("--verify supports the one-way mode")
+    if args.verify and (args.duplex or args.pairs > 1):
+        ap.error("--verify supports the plain one-way mode")
 
     ctx = mp.get_context("spawn")
     a, b = ctx.Pipe()
@@ -297,6 +408,9 @@ def main():
                          args=(a, args, q, True, ns_child))
         ps = ctx.Process(target=_duplex_worker,
                          args=(b, sender_args, q, False))
+    elif args.pairs > 1:
+        pr = ctx.Process(target=_receiver_pairs, args=(a, args, q, ns_child))
+        ps = ctx.Process(target=_sender_pairs, args=(b, sender_args, q))
     else:
         pr = ctx.Process(target=_receiver, args=(a, args, q, ns_child))
         ps = ctx.Process(target=_sender, args=(b, sender_args, q))
