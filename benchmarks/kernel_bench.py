@@ -53,7 +53,7 @@ def main():
     src = torch.randn(16 << 20, device=dev)  # 64 MiB
     dst = torch.empty_like(src)
     timed("copy_bytes_d2d",
-          lambda: ops.copy_bytes(dst, src), 2 * src.nbytes())
+          lambda: ops.copy_bytes(dst, src), 2 * src.nbytes)
 
 
 if __name__ == "__main__":
