@@ -80,6 +80,20 @@ def main():
     def poll():
         nonlocal completed, bytes_total, last_progress
         for m in list(live):
+            if m.get("group"):
+                if not m["sdone"]:
+                    m["sdone"] = all(p.test(s)[0] for s in m["sreqs"])
+                if not m["rdone"]:
+                    m["rdone"], szs = p.test_n(m["rreq"], m["n"])
+                    if m["rdone"]:
+                        assert szs == m["gsizes"], (szs, m["gsizes"])
+                if m["sdone"] and m["rdone"]:
+                    m["verify"]()
+                    live.remove(m)
+                    completed += m["n"]
+                    bytes_total += m["size"]
+                    last_progress = time.monotonic()
+                continue
             if not m["sdone"]:
                 m["sdone"], _ = p.test(m["sreq"])
             if not m["rdone"]:
@@ -133,6 +147,40 @@ def main():
                 live.append(dict(sreq=sreq, rreq=rreq, sdone=False,
                                  rdone=False, verify=verify, size=nbytes,
                                  gpu=True))
+            elif rng.random() < 0.2:
+                # grouped receive: 2-4 messages posted as ONE request
+                n = rng.randint(2, 4)
+                gsizes = [rng.choice(SIZES) for _ in range(n)]
+                payloads = [(bytes([rng.getrandbits(8)]) * s) for s in
+                            gsizes]
+                sbufs = [C.create_string_buffer(pl, max(s, 1))
+                         for pl, s in zip(payloads, gsizes)]
+                rbufs = [C.create_string_buffer(s + 1) for s in gsizes]
+                tags = [rng.randrange(1 << 16) for _ in range(n)]
+                greq, _ = p.irecv_n(rcomm, rbufs, gsizes, rmh, tags)
+                if greq is None:  # slot backpressure
+                    poll()
+                    break
+                sreqs = []
+                t0 = time.monotonic()
+                for sb, s, t in zip(sbufs, gsizes, tags):
+                    sr = p.isend(scomm, sb, s, smh, tag=t)
+                    while sr is None:
+                        poll()
+                        sr = p.isend(scomm, sb, s, smh, tag=t)
+                        if time.monotonic() - t0 > 30:
+                            dump_state(p, scomm, rcomm, live)
+                    sreqs.append(sr)
+
+                def verify(payloads=payloads, rbufs=rbufs, gsizes=gsizes):
+                    for pl, rb, s in zip(payloads, rbufs, gsizes):
+                        assert rb.raw[:s] == pl, "grouped payload corrupt"
+
+                live.append(dict(group=True, sreqs=sreqs, rreq=greq, n=n,
+                                 gsizes=gsizes, verify=verify,
+                                 size=sum(gsizes), sdone=False,
+                                 rdone=False, gpu=False,
+                                 _bufs=(sbufs, rbufs)))  # keep alive
             else:
                 payload = (bytes([rng.getrandbits(8)]) * size) if size \
                     else b""
