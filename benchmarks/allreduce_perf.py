@@ -53,6 +53,11 @@ def main():
                 os.environ[k] = f"{v}:{os.environ.get(k, '')}".rstrip(":")
             else:
                 os.environ.setdefault(k, v)
+        if args.force_net:
+            # pin OUR net by name: if the plugin failed to load, RCCL must
+            # error out rather than silently falling back to its internal
+            # socket transport (which would fake a 1.0 plugin/stock ratio)
+            os.environ.setdefault("NCCL_NET", "BaguaNetAMD")
     elif args.force_net:
         os.environ["NCCL_P2P_DISABLE"] = "1"
         os.environ["NCCL_SHM_DISABLE"] = "1"
