@@ -240,7 +240,12 @@ struct TcpSock {
   // io_uring engine per-socket state (unused by the epoll engine)
   static constexpr int kUrBatch = 8;  // chunks per WRITEV submission
   struct {
-    uint8_t op = 0;  // 0 none, 1 send(writev), 2 recv
+    // 0 none, 1 send(writev), 2 recv.  Atomic because the proxy's inline
+    // send reads it under owner_busy while the engine thread transitions
+    // it: an armed op (op==1) also serves as the CQE handler's mutual
+    // exclusion — the proxy bails whenever op != 0, so the handler may
+    // mutate ur.* freely until it clears op (release) as its last step.
+    std::atomic<uint8_t> op{0};
     bool closing = false;  // removal in progress: no resubmission
     bool eof = false;      // recv: orderly peer shutdown — never re-arm
     // batched send: up to kUrBatch chunks in one ordered WRITEV

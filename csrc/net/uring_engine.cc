@@ -41,6 +41,11 @@ namespace baguanet {
 
 namespace {
 
+// set by submit_send(allow_ring=false) when a batch tail needs a ring op
+// the calling (proxy) thread cannot arm; thread-local because several
+// proxy threads may inline concurrently
+thread_local bool ring_needed_ = false;
+
 int sys_uring_setup(unsigned entries, io_uring_params* p) {
   return (int)syscall(__NR_io_uring_setup, entries, p);
 }
@@ -208,7 +213,41 @@ class UringIoThread : public IIoThread {
   // interleave bytes, and a single chunk per op starves the kernel
   // between completion and resubmission (measured 24 vs 47 GB/s against
   // the epoll engine before batching).
-  void submit_send(TcpSock* s) {
+  //
+  // owner_busy exclusion (same protocol as the epoll engine): every
+  // send-side entry — ADD/KICK tasks, the spin retry loop, the CQE
+  // handler's resubmission, and the PROXY's inline attempt — goes
+  // through submit_send_gated / try_inline_send.  The proxy must never
+  // touch the ring, so its path runs submit_send(allow_ring=false):
+  // an EAGAIN tail marks ring_needed_ and falls back to a kick so the
+  // engine thread arms the WRITEV.
+  void submit_send_gated(TcpSock* s) {
+    if (s->owner_busy.exchange(1, std::memory_order_acquire) != 0) {
+      s->rekick.store(true, std::memory_order_release);
+      return;
+    }
+    submit_send(s);
+    s->owner_busy.store(0, std::memory_order_release);
+    if (s->rekick.exchange(false, std::memory_order_acq_rel)) kick(s);
+  }
+
+  bool try_inline_send(TcpSock* s) override {
+    if (s->fd < 0) return false;
+    if (s->owner_busy.exchange(1, std::memory_order_acquire) != 0)
+      return false;  // engine thread owns it right now — caller kicks
+    bool ok = false;
+    if (s->ur.op == 0 && !s->ur.closing &&
+        !s->scomm->error.load(std::memory_order_relaxed)) {
+      ring_needed_ = false;
+      submit_send(s, /*allow_ring=*/false);
+      ok = !ring_needed_;  // partial batch: owner must arm the WRITEV
+    }
+    s->owner_busy.store(0, std::memory_order_release);
+    if (s->rekick.exchange(false, std::memory_order_acq_rel)) kick(s);
+    return ok;
+  }
+
+  void submit_send(TcpSock* s, bool allow_ring = true) {
     SendComm* c = s->scomm;
     if (s->ur.op || s->ur.closing ||
         c->error.load(std::memory_order_relaxed))
@@ -274,17 +313,21 @@ class UringIoThread : public IIoThread {
       s->ur.done += (uint32_t)n;
       if (s->ur.done == s->ur.batch_bytes) {
         finish_batch(s);
-        submit_send(s);  // claim + send the next batch inline too
+        submit_send(s, allow_ring);  // claim + send the next batch too
         return;
       }
       // partial: rebuild the iovec for the unwritten tail
-      submit_send_tail(s);
+      submit_send_tail(s, allow_ring);
       return;
     }
     if (n < 0 && errno != EAGAIN && errno != EWOULDBLOCK &&
         errno != EINTR) {
       s->scomm->error.store(errno);
       BNET_WARN("bnet(uring) inline send error: %s", strerror(errno));
+      return;
+    }
+    if (!allow_ring) {
+      ring_needed_ = true;  // proxy path: the engine thread must arm
       return;
     }
     arm_writev(s, iovn);
@@ -304,7 +347,7 @@ class UringIoThread : public IIoThread {
   }
 
   // rebuild the iov for the unwritten tail of the batch and arm it
-  void submit_send_tail(TcpSock* s) {
+  void submit_send_tail(TcpSock* s, bool allow_ring = true) {
     int iovn = 0;
     uint32_t skip = s->ur.done;
     for (int i = 0; i < s->ur.nchunks; i++) {
@@ -325,6 +368,10 @@ class UringIoThread : public IIoThread {
     }
     if (iovn == 0) {
       finish_batch(s);
+      return;
+    }
+    if (!allow_ring) {
+      ring_needed_ = true;
       return;
     }
     arm_writev(s, iovn);
@@ -348,11 +395,14 @@ class UringIoThread : public IIoThread {
   }
 
   void on_send_cqe(TcpSock* s, int res) {
-    s->ur.op = 0;
+    // ur.* is mutated while ur.op is STILL 1 — the proxy's inline path
+    // bails on op != 0, so the armed op doubles as the CQE handler's
+    // exclusion; op clears (release) only after the state is consistent
     if (res <= 0) {
+      s->ur.op.store(0, std::memory_order_release);
       if (res == -ECANCELED) return;  // teardown cancel
       if (res == -EAGAIN || res == -EINTR) {
-        submit_send(s);  // retry
+        submit_send_gated(s);  // retry
         return;
       }
       s->scomm->error.store(res ? -res : EPIPE);
@@ -361,7 +411,8 @@ class UringIoThread : public IIoThread {
     }
     s->ur.done += (uint32_t)res;
     if (s->ur.done == s->ur.batch_bytes) finish_batch(s);
-    submit_send(s);  // remainder / next batch
+    s->ur.op.store(0, std::memory_order_release);
+    submit_send_gated(s);  // remainder / next batch
   }
 
   // ---- recv path -------------------------------------------------------
@@ -447,7 +498,7 @@ class UringIoThread : public IIoThread {
           if (t.s->is_recv)
             submit_recv(t.s);
           else
-            submit_send(t.s);
+            submit_send_gated(t.s);
           break;
         case Task::REMOVE: {
           TcpSock* s = t.s;
@@ -486,7 +537,7 @@ class UringIoThread : public IIoThread {
               else
                 submit_recv(t.s);  // no-op if an op is already in flight
             } else {
-              submit_send(t.s);
+              submit_send_gated(t.s);
             }
           }
           break;
@@ -555,7 +606,7 @@ class UringIoThread : public IIoThread {
             if (!s->ur.eof && !s->parked.load(std::memory_order_relaxed))
               submit_recv(s);
           } else {
-            submit_send(s);
+            submit_send_gated(s);
           }
         }
         ring_.enter(0);
