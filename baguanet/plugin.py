@@ -307,21 +307,21 @@ class Plugin:
         self._check(self.vt.closeListen(lcomm), "closeListen")
 
 
-_preloaded = None
+_preloaded = {}
 
 
-def preload() -> None:
+def preload(name: str = "bagua") -> None:
     """dlopen the plugin with its SONAME registered, so RCCL's later
-    dlopen("librccl-net-bagua.so") resolves to the already-loaded library.
+    dlopen("librccl-net-<name>.so") resolves to the already-loaded
+    library.  ``name="bagua6"`` preloads the v6-only escape hatch.
 
     Needed because mutating LD_LIBRARY_PATH inside a running process does
     not affect dlopen search paths (glibc snapshots it at startup) — the
     torchrun-launched bench sets env in-process.
     """
-    global _preloaded
-    if _preloaded is None:
-        _preloaded = C.CDLL(str(PLUGIN_DIR / "librccl-net-bagua.so"),
-                            mode=C.RTLD_GLOBAL)
+    if name not in _preloaded:
+        _preloaded[name] = C.CDLL(str(PLUGIN_DIR / f"librccl-net-{name}.so"),
+                                  mode=C.RTLD_GLOBAL)
 
 
 def rccl_env(
