@@ -135,3 +135,27 @@ def test_cross_engine_interop():
     length frames differed)."""
     _run_pair(send_engine="EPOLL", recv_engine="URING")
     _run_pair(send_engine="URING", recv_engine="EPOLL")
+
+
+def test_ring_allreduce_cpu():
+    """Verified ring all-reduce over the plugin at 3 ranks (odd count
+    exercises non-power-of-two chunking)."""
+    import json
+    import subprocess
+    import sys
+
+    REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+    res = subprocess.run(
+        [sys.executable,
+         os.path.join(REPO, "benchmarks", "ring_allreduce.py"),
+         "--ranks", "3", "--sizes", "999936", "--iters", "2",
+         "--warmup", "1", "--json"],
+        capture_output=True, text=True, timeout=300, cwd=REPO,
+        env={**os.environ, "BNET_IO_THREADS": "2"},
+    )
+    assert res.returncode == 0, res.stderr[-2000:]
+    out = json.loads([l for l in res.stdout.splitlines()
+                      if l.startswith("{")][-1])
+    assert out["verified"] and out["ranks"] == 3
+    assert out["results"][0]["busbw_GBps"] > 0
