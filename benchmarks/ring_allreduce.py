@@ -28,9 +28,23 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import numpy as np  # noqa: E402
 
 
-def _rank(rank, nranks, prev_conn, next_conn, args, out_q):
+def _rank(rank, nranks, prev_conn, next_conn, args, out_q, ns_conn=None):
     """prev_conn talks to rank-1 (we receive from it), next_conn to
     rank+1 (we send to it)."""
+    if ns_conn is not None:  # --shaped, rank 1 of 2: own netns behind TBF
+        sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+        import netns_rig as rig
+
+        rig.unshare_newnet()
+        ns_conn.send(os.getpid())
+        assert ns_conn.recv() == "veth-moved"
+        rig.child_setup(args.shaped, 10.0)
+        os.environ["NCCL_SOCKET_IFNAME"] = rig.CHILD_IF
+    elif args.shaped is not None:  # rank 0 stays in the parent ns
+        sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+        import netns_rig as rig
+
+        os.environ["NCCL_SOCKET_IFNAME"] = rig.PARENT_IF
     os.environ.setdefault("NCCL_SOCKET_IFNAME", "lo")
     os.environ.setdefault("BNET_MIN_CHUNKSIZE", "131072")
     from baguanet.plugin import Plugin
@@ -172,27 +186,54 @@ def main():
     ap.add_argument("--iters", type=int, default=10)
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--json", action="store_true")
+    ap.add_argument("--shaped", type=float, default=None,
+                    help="2 ranks only: rank 1 runs in its own netns "
+                         "behind a veth pair TBF-shaped to this many "
+                         "Gbit/s — a two-node-over-limited-TCP analogue "
+                         "(BASELINE config 2).  Needs CAP_NET_ADMIN")
     args = ap.parse_args()
+    if args.shaped is not None and args.ranks != 2:
+        ap.error("--shaped supports exactly 2 ranks")
 
     ctx = mp.get_context("spawn")
     # pipe ring: conn[r] connects rank r (as next_conn) with r+1 (as prev)
     pipes = [ctx.Pipe() for _ in range(args.ranks)]
     q = ctx.Queue()
+    ns_parent = ns_child = None
+    if args.shaped is not None:
+        ns_parent, ns_child = ctx.Pipe()
     procs = []
     for r in range(args.ranks):
         prev_conn = pipes[(r - 1) % args.ranks][1]  # to rank r-1
         next_conn = pipes[r][0]                     # to rank r+1
         procs.append(ctx.Process(
             target=_rank, args=(r, args.ranks, prev_conn, next_conn,
-                                args, q)))
-    for pr in procs:
-        pr.start()
+                                args, q,
+                                ns_child if r == 1 else None)))
+    procs[1 if args.shaped is not None else 0].start()
+    if args.shaped is not None:
+        sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+        import netns_rig as rig
+
+        child_pid = ns_parent.recv()
+        rig.parent_setup(child_pid)
+        if args.shaped > 0:
+            rig.parent_shape(args.shaped, 10.0)
+        ns_parent.send("veth-moved")
+        procs[0].start()
+    else:
+        for pr in procs[1:]:
+            pr.start()
     outs = {}
     for _ in range(args.ranks):
         r, res = q.get(timeout=900)
         outs[r] = res
     for pr in procs:
         pr.join(30)
+    if args.shaped is not None:
+        import netns_rig as rig
+
+        rig.parent_teardown()
     # max-over-ranks per size (nccl-tests reports the slowest rank)
     merged = []
     for i, size_res in enumerate(outs[0]):
