@@ -750,3 +750,28 @@ def test_rank_templated_telemetry_files(tmp_path):
         "BNET_METRICS_FILE": templ,
     }) == "ok"
     assert (tmp_path / "rank-5.prom").exists()
+
+
+def _pinned_stats(env, q):
+    for k, v in env.items():
+        os.environ[k] = v
+    import ctypes as C
+
+    from baguanet.plugin import Plugin
+
+    p = Plugin()
+    used = C.c_size_t(1)
+    budget = C.c_size_t(0)
+    p.lib.bnet_pinned_stats(C.byref(used), C.byref(budget))
+    q.put((used.value, budget.value))
+
+
+def test_pinned_budget_knob():
+    """BNET_PINNED_BUDGET reaches the staging accountant (no GPU needed:
+    nothing is pinned yet, but the budget must reflect the env)."""
+    used, budget = _run_sub(_pinned_stats, {
+        "NCCL_SOCKET_IFNAME": "lo",
+        "BNET_PINNED_BUDGET": str(123 << 20),
+    })
+    assert used == 0
+    assert budget == 123 << 20

@@ -47,3 +47,22 @@ def test_shaped_p2p_saturates_and_respects_ceiling():
     line_rate = 10 / 8  # GB/s
     assert bw >= 0.8 * line_rate, f"only {bw} GB/s on a 10 Gb shape"
     assert bw <= 1.2 * line_rate, f"{bw} GB/s exceeds the 10 Gb ceiling"
+
+
+def test_shaped_ring_allreduce_two_nodes():
+    """The 2-'node' ring all-reduce analogue of BASELINE config 2: result
+    verified and busbw bounded by the 5 Gb shape."""
+    res = subprocess.run(
+        [sys.executable,
+         os.path.join(REPO, "benchmarks", "ring_allreduce.py"),
+         "--ranks", "2", "--shaped", "5", "--sizes", "4194304",
+         "--iters", "2", "--warmup", "1", "--json"],
+        capture_output=True, text=True, timeout=300, cwd=REPO,
+        env={**os.environ, "BNET_IO_THREADS": "2"},
+    )
+    assert res.returncode == 0, res.stderr[-2000:]
+    out = json.loads([l for l in res.stdout.splitlines()
+                      if l.startswith("{")][-1])
+    assert out["verified"]
+    bw = out["results"][0]["busbw_GBps"]
+    assert 0 < bw <= 5 / 8 * 1.2, bw  # under the shaped ceiling
